@@ -1,0 +1,92 @@
+"""In-tree build of starway_amd._core with hipcc (gfx950).
+
+Used by setup.py, __graft_entry__.build() and tests. Compiles each TU to
+build/*.o with mtime-based incrementality, then links the extension .so into
+starway_amd/ so it travels with the repo snapshot to GPU boxes.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import sysconfig
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+CSRC = REPO / "csrc"
+BUILD = REPO / "build"
+PKG = REPO / "starway_amd"
+
+SOURCES = ["engine.cpp", "gpu.cpp", "module.cpp", "kernels.hip"]
+
+HIPCC = os.environ.get("STARWAY_HIPCC", "hipcc")
+ARCH = os.environ.get("STARWAY_OFFLOAD_ARCH", "gfx950")
+
+
+def _pybind11_include() -> str:
+    import pybind11
+
+    return pybind11.get_include()
+
+
+def ext_path() -> Path:
+    suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+    return PKG / f"_core{suffix}"
+
+
+def _needs_rebuild(obj: Path, src: Path, hdrs: list[Path]) -> bool:
+    if not obj.exists():
+        return True
+    omt = obj.stat().st_mtime
+    if src.stat().st_mtime > omt:
+        return True
+    return any(h.stat().st_mtime > omt for h in hdrs)
+
+
+def build(verbose: bool = True, debug: bool = False) -> Path:
+    BUILD.mkdir(exist_ok=True)
+    hdrs = sorted(CSRC.glob("*.hpp"))
+    py_inc = sysconfig.get_paths()["include"]
+    common = [
+        "-O3",
+        "-std=c++20",
+        f"--offload-arch={ARCH}",
+        "-fPIC",
+        "-I",
+        str(CSRC),
+        "-I",
+        py_inc,
+        "-I",
+        _pybind11_include(),
+        "-Wno-unused-result",
+    ]
+    if debug:
+        common += ["-g", "-DSW_DEBUG"]
+    else:
+        common += ["-DNDEBUG"]
+
+    objs: list[Path] = []
+    linked_any = False
+    for name in SOURCES:
+        src = CSRC / name
+        obj = BUILD / (name.replace(".", "_") + ".o")
+        objs.append(obj)
+        if _needs_rebuild(obj, src, hdrs):
+            cmd = [HIPCC, *common, "-c", str(src), "-o", str(obj)]
+            if verbose:
+                print("[build_ext]", " ".join(cmd), flush=True)
+            subprocess.run(cmd, check=True)
+            linked_any = True
+
+    out = ext_path()
+    if linked_any or not out.exists():
+        cmd = [HIPCC, "-shared", "-fPIC", *[str(o) for o in objs], "-o", str(out)]
+        if verbose:
+            print("[build_ext]", " ".join(cmd), flush=True)
+        subprocess.run(cmd, check=True)
+    return out
+
+
+if __name__ == "__main__":
+    build(debug="--debug" in sys.argv)
+    print(f"built {ext_path()}")

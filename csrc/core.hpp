@@ -1,0 +1,415 @@
+// starway_amd core — MI355X-native tagged async zero-copy messaging engine.
+//
+// Brand-new design with the capability surface of Clouder0/starway
+// (reference: /root/reference/src/bindings/main.hpp — API contract only; the
+// transport below is our own: TCP control/CPU-data plane + hipIpc/xGMI GPU
+// data plane with hand-written gfx950 copy kernels, no UCX).
+//
+// Architecture invariants (mirrors reference main.cpp:150-161 design stance):
+//  * all transport state is owned by ONE progress thread per Client/Server
+//  * the Python thread communicates with it only through a command queue and
+//    atomic status flags
+//  * Python callbacks fire only from controlled points with the GIL held
+#pragma once
+
+#include <arpa/inet.h>
+#include <atomic>
+#include <cstdint>
+#include <cstring>
+#include <deque>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <set>
+#include <string>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+#include <pybind11/pybind11.h>
+
+namespace py = pybind11;
+
+namespace sw {
+
+// ---------------------------------------------------------------------------
+// Wire protocol (little-endian, x86-64 only)
+// ---------------------------------------------------------------------------
+
+constexpr uint32_t kMagic = 0x53574159;  // "SWAY"
+constexpr uint32_t kProtoVersion = 1;
+
+enum FrameType : uint16_t {
+  FT_HELLO = 1,      // payload: PeerInfo blob (both directions on connect)
+  FT_EAGER = 2,      // payload: message bytes; hdr.tag/size = message tag/len
+  FT_RTS = 3,        // payload: RtsDesc — GPU rendezvous "ready to send"
+  FT_RECV_DONE = 4,  // hdr.op_id = sender-side op id whose data was pulled
+  FT_RECV_FAIL = 5,  // hdr.op_id + payload = utf-8 error string
+  FT_FLUSH_REQ = 6,  // hdr.op_id = flush id; receiver echoes FLUSH_ACK
+  FT_FLUSH_ACK = 7,  // hdr.op_id echo
+  FT_BYE = 8,        // graceful close notification
+};
+
+#pragma pack(push, 1)
+struct FrameHeader {
+  uint32_t magic;
+  uint16_t type;
+  uint16_t flags;
+  uint64_t tag;
+  uint64_t size;   // payload bytes that follow this header
+  uint64_t op_id;  // sender-side op id (RTS/RECV_DONE/FLUSH correlation)
+  uint64_t aux;    // type-specific (EAGER: full message length)
+};
+#pragma pack(pop)
+static_assert(sizeof(FrameHeader) == 40);
+
+constexpr int kIpcHandleBytes = 64;  // HIP_IPC_HANDLE_SIZE
+
+// GPU rendezvous descriptor — carried as RTS payload. The receiver pulls the
+// data over xGMI (peer kernel copy) or locally, then acks with RECV_DONE.
+#pragma pack(push, 1)
+struct RtsDesc {
+  uint8_t src_uuid[16];  // sender process uuid — same-process => use raw ptr
+  int32_t device;        // sender device ordinal
+  uint8_t use_ipc;       // 1 => ipc_handle+offset valid, else raw ptr
+  uint8_t pad[3];
+  uint8_t ipc_handle[kIpcHandleBytes];
+  uint64_t offset;   // byte offset of message start within the ipc allocation
+  uint64_t raw_ptr;  // device pointer (same-process fast path)
+};
+#pragma pack(pop)
+
+// PeerInfo blob (HELLO payload / worker-address blob body):
+//   u32 version | u64 pid | 16B uuid | u8 has_gpu | i32 gpu_count |
+//   u16 name_len | name bytes
+struct PeerInfo {
+  uint64_t pid = 0;
+  uint8_t uuid[16] = {0};
+  bool has_gpu = false;
+  int32_t gpu_count = 0;
+  std::string name;
+};
+
+std::vector<uint8_t> encode_peer_info(const PeerInfo& pi);
+bool decode_peer_info(const uint8_t* data, size_t len, PeerInfo* out);
+const uint8_t* process_uuid();  // 16 bytes, stable for this process
+
+// ---------------------------------------------------------------------------
+// Buffers
+// ---------------------------------------------------------------------------
+
+// A caller-provided message buffer. device < 0 => host memory.
+struct BufferRef {
+  uint8_t* ptr = nullptr;
+  size_t size = 0;
+  int device = -1;
+};
+
+// Uninitialized heap buffer (std::vector zero-fills on resize, which stalls
+// the progress thread for multi-GiB unexpected-message staging).
+struct RawBuf {
+  std::unique_ptr<uint8_t[]> p;
+  size_t n = 0;
+  void alloc(size_t s) {
+    p.reset(new uint8_t[s]);  // default-init: uninitialized for uint8_t
+    n = s;
+  }
+  uint8_t* data() { return p.get(); }
+  size_t size() const { return n; }
+  bool empty() const { return n == 0; }
+  void clear() {
+    p.reset();
+    n = 0;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// Ops
+// ---------------------------------------------------------------------------
+
+enum class OpType : uint8_t { Send, Recv, Flush, FlushEp, Connect };
+
+struct Connection;    // fwd
+struct EndpointInfo;  // fwd
+
+struct Op {
+  uint64_t id = 0;
+  OpType type = OpType::Send;
+  BufferRef buf;
+  uint64_t tag = 0;
+  uint64_t tag_mask = 0;
+  Connection* conn = nullptr;  // send target / recv source (once matched)
+
+  // Python state — touched only with the GIL held.
+  py::object done_cb;   // Send/Flush: (); Recv: (sender_tag, length)
+  py::object fail_cb;   // (reason: str)
+  py::object keepalive; // the user's numpy array / torch tensor
+
+  // Keeps the ServerEndpoint alive across the command-queue hop; the engine
+  // re-resolves conn from it at processing time.
+  std::shared_ptr<EndpointInfo> ep_ref;
+
+  // Progress state (engine thread only).
+  uint64_t recv_len = 0;          // actual message length for recv completion
+  uint64_t recv_sender_tag = 0;
+  bool gpu_send_awaiting_ack = false;
+  // Flush bookkeeping:
+  std::set<std::pair<Connection*, uint64_t>> flush_acks_pending;
+  std::set<uint64_t> flush_ops_pending;  // GPU send op ids in flight
+};
+
+// ---------------------------------------------------------------------------
+// Endpoint info (exposed to Python as ServerEndpoint)
+// ---------------------------------------------------------------------------
+
+struct EndpointInfo {
+  std::string name;
+  std::string local_addr;
+  int local_port = 0;
+  std::string remote_addr;
+  int remote_port = 0;
+  std::vector<std::pair<std::string, std::string>> transports;
+  Connection* conn = nullptr;  // nulled when the connection dies
+  class Engine* owner = nullptr;
+};
+
+// ---------------------------------------------------------------------------
+// Connection — one TCP stream + its rx/tx state (engine thread only)
+// ---------------------------------------------------------------------------
+
+struct TxItem {
+  std::vector<uint8_t> head;      // header (+ inline payload for small frames)
+  const uint8_t* ext = nullptr;   // zero-copy user payload (may be null)
+  size_t ext_len = 0;
+  bool has_keepalive = false;
+  bool is_data = false;           // EAGER/RTS frames: droppable on close
+  py::object keepalive;           // dropped (under GIL) once fully written
+};
+
+struct UnexpectedMsg {
+  uint64_t tag = 0;
+  uint64_t size = 0;
+  Connection* conn = nullptr;
+  bool is_rts = false;
+  uint64_t sender_op_id = 0;
+  RtsDesc rts{};
+  RawBuf data;  // eager staging (uninitialized alloc)
+  size_t got = 0;
+  bool complete = false;
+  Op* bound_recv = nullptr;   // recv matched while message still streaming in
+};
+
+struct Connection {
+  int fd = -1;
+  uint64_t conn_id = 0;
+  std::string local_addr, remote_addr;
+  int local_port = 0, remote_port = 0;
+  bool hello_sent = false;
+  bool hello_received = false;
+  PeerInfo peer;
+  bool dead = false;
+  std::shared_ptr<EndpointInfo> ep;  // server side
+
+  // --- rx parser ---
+  enum class RxState { Header, Payload } rx_state = RxState::Header;
+  FrameHeader rx_hdr{};
+  size_t rx_got = 0;                 // bytes of current header/payload read
+  std::vector<uint8_t> rx_small;     // staging for small control payloads
+  // Eager delivery target (exactly one of these while streaming a message):
+  Op* rx_recv_op = nullptr;          // matched posted recv (zero-copy fill)
+  UnexpectedMsg* rx_unexp = nullptr; // unmatched (staged)
+  std::vector<uint8_t> rx_discard;   // truncation sink
+  uint64_t rx_msg_remaining = 0;
+  bool rx_truncated = false;
+  // GPU eager-recv host bounce (posted recv buffer is on device):
+  RawBuf rx_gpu_bounce;
+
+  // --- tx ---
+  std::deque<TxItem> txq;
+  size_t tx_front_written = 0;
+  size_t tx_bytes_queued = 0;
+
+  bool want_write() const { return !txq.empty(); }
+};
+
+// ---------------------------------------------------------------------------
+// Completion record — Python callback to run under the GIL
+// ---------------------------------------------------------------------------
+
+struct Completion {
+  enum class Kind : uint8_t { SendDone, RecvDone, FlushDone, Fail, Connect,
+                              Accept, Close } kind;
+  Op* op = nullptr;                  // owned; deleted after firing
+  std::string error;                 // Fail / Connect
+  uint64_t a = 0, b = 0;             // RecvDone: (sender_tag, length)
+  py::object cb0;                    // Connect/Accept/Close callback
+  std::shared_ptr<EndpointInfo> ep;  // Accept
+};
+
+// ---------------------------------------------------------------------------
+// Engine — shared core of Client and Server
+// ---------------------------------------------------------------------------
+
+class Context {
+ public:
+  Context() = default;
+  Context(const Context&) = delete;
+  Context& operator=(const Context&) = delete;
+};
+
+// status: 0 void / 1 init / 2 running / 3 closing / 4 closed
+// (state machine contract of reference main.hpp:173-174)
+class Engine {
+ public:
+  enum Mode { ClientMode, ServerMode };
+
+  explicit Engine(Mode mode);
+  ~Engine();
+
+  // ---- Python-thread API (GIL held on entry) ----
+  void listen(const std::string& addr, int port);     // server, raises
+  std::vector<uint8_t> listen_address();              // server worker mode
+  void connect(const std::string& addr, int port, py::object cb);
+  void connect_address(const std::vector<uint8_t>& blob, py::object cb);
+  std::vector<uint8_t> get_worker_address();
+  void set_accept_callback(py::object cb);
+  void close(py::object cb);
+
+  void send(std::shared_ptr<EndpointInfo> ep, BufferRef buf, uint64_t tag,
+            py::object done, py::object fail, py::object keepalive);
+  void recv(BufferRef buf, uint64_t tag, uint64_t mask, py::object done,
+            py::object fail, py::object keepalive);
+  void flush(py::object done, py::object fail);
+  void flush_ep(std::shared_ptr<EndpointInfo> ep, py::object done,
+                py::object fail);
+  std::vector<std::shared_ptr<EndpointInfo>> list_clients();
+  double evaluate_perf(std::shared_ptr<EndpointInfo> ep, uint64_t msg_size);
+
+  int status() const { return status_.load(std::memory_order_acquire); }
+
+ private:
+  // ---- engine thread ----
+  void thread_main();
+  void loop_iteration(bool& did_work);
+  void drain_commands(std::vector<Op*>& cmds);
+  void process_command(Op* op);
+  void do_connect_start();
+  void poll_sockets(int timeout_ms, bool& did_work);
+  void handle_readable(Connection* c, bool& did_work);
+  void handle_writable(Connection* c, bool& did_work);
+  void accept_new(bool& did_work);
+  void on_frame(Connection* c);           // full header parsed, size==0 or
+  void on_frame_payload(Connection* c);   // payload fully staged in rx_small
+  void begin_eager(Connection* c);
+  void finish_eager_into_recv(Connection* c);
+  void handle_rts(Connection* c, const RtsDesc& rts, uint64_t tag,
+                  uint64_t size, uint64_t sender_op);
+  void start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
+                      uint64_t size, uint64_t sender_op, Connection* c);
+  void match_or_stash_recv(Op* op);
+  bool try_match_unexpected(Op* op);
+  void complete_recv_from_unexpected(Op* op, UnexpectedMsg* um);
+  void poll_gpu(bool& did_work);
+  void enqueue_frame(Connection* c, FrameType t, uint64_t tag, uint64_t op_id,
+                     uint64_t aux, const void* payload, size_t payload_len,
+                     bool priority);
+  void enqueue_eager(Connection* c, Op* op);
+  void send_hello(Connection* c);
+  void on_hello(Connection* c);
+  void on_conn_dead(Connection* c);
+  void check_flush_progress(Connection* c, uint64_t acked_flush_id);
+  void on_gpu_send_acked(uint64_t op_id, bool failed, const std::string& err);
+  void teardown();
+  void fire_completions();  // acquires GIL, drains completions_
+  void complete(Completion&& comp);
+  void fail_op(Op* op, const std::string& reason);
+  void wake();
+  Connection* make_listener(const std::string& addr, int port);
+
+  double perf_model(Connection* c, uint64_t msg_size) const;
+
+ public:
+  Mode mode_;
+  std::atomic<int> status_{0};
+
+ private:
+  // Command queue: Python threads -> engine thread.
+  std::mutex cmd_mu_;
+  std::vector<Op*> cmd_queue_;
+  std::atomic<bool> cmd_pending_{false};
+  int wake_fds_[2] = {-1, -1};  // self-pipe to interrupt poll()
+
+  std::thread thread_;
+  std::atomic<uint64_t> next_op_id_{1};
+
+  // Engine-thread state.
+  int listen_fd_ = -1;
+  std::string listen_host_;
+  int listen_port_ = 0;
+  bool worker_mode_ = false;
+  std::vector<std::unique_ptr<Connection>> conns_;
+  uint64_t next_conn_id_ = 1;
+  std::deque<Op*> posted_recvs_;
+  std::deque<std::unique_ptr<UnexpectedMsg>> unexpected_;
+  std::vector<Op*> pending_flushes_;
+  std::unordered_map<uint64_t, Op*> gpu_sends_;  // op id -> awaiting ack
+  struct GpuPull;  // defined in engine.cpp (holds hipEvent)
+  std::vector<std::unique_ptr<GpuPull>> gpu_pulls_;
+  uint64_t next_flush_id_ = 1;
+
+  // Connect state (client).
+  std::string connect_host_;
+  int connect_port_ = 0;
+  py::object connect_cb_;
+  bool connect_requested_ = false;
+  bool connect_done_ = false;
+
+  py::object accept_cb_;
+  py::object close_cb_;
+  bool have_accept_cb_ = false;
+
+  std::vector<Completion> completions_;
+  // py::objects whose final decref must wait for a GIL hold (moved here from
+  // engine-thread contexts; cleared inside fire_completions).
+  std::vector<py::object> dead_objs_;
+
+  // Endpoint registry (shared with Python thread under ep_mu_).
+  std::mutex ep_mu_;
+  std::vector<std::shared_ptr<EndpointInfo>> eps_;
+
+  uint64_t idle_iters_ = 0;
+};
+
+// ---------------------------------------------------------------------------
+// GPU layer (gpu.cpp) — all HIP calls live behind this interface so the
+// engine compiles and runs on GPU-less hosts.
+// ---------------------------------------------------------------------------
+
+namespace gpu {
+bool available();
+int device_count();
+// Fill an RtsDesc for a device buffer (ipc handle or raw ptr).
+bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err);
+// Begin an async pull of `size` bytes described by `rts` into dst (host or
+// device). Returns an opaque ticket (hipEvent) or null on error.
+void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
+                 std::string* err);
+// Begin an async host->device upload into a device buffer. The host memory
+// must stay valid until the ticket completes; hand its ownership to the
+// ticket with attach_bounce.
+void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
+                std::string* err);
+void attach_bounce(void* ticket, RawBuf&& bounce);
+// Poll a ticket: 1 done, 0 pending, -1 error.
+int poll_ticket(void* ticket, std::string* err);
+void free_ticket(void* ticket);
+void synchronize_all();
+bool copy_device_sync(void* dst, const void* src, size_t n, int device,
+                      std::string* err);
+// Stats for evaluate_perf / transports introspection.
+double same_gpu_copy_gbps();
+double xgmi_link_gbps();
+}  // namespace gpu
+
+}  // namespace sw

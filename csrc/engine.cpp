@@ -1,0 +1,1612 @@
+// starway_amd engine — progress thread, TCP plane, tag matching, GPU
+// rendezvous orchestration.
+//
+// Design notes (API/behavior parity targets cite the reference):
+//  * one progress thread per Client/Server object; all transport state is
+//    thread-confined (reference src/bindings/main.cpp:234-550 invariant)
+//  * status machine 0 void / 1 init / 2 running / 3 closing / 4 closed
+//    (reference src/bindings/main.hpp:173-174)
+//  * tag matching: recv (tag, mask) matches message m iff
+//    (m.tag & mask) == (tag & mask); posted recvs FIFO, unexpected messages
+//    kept in arrival order (this replaces what UCX's matching engine did for
+//    the reference at ucp_tag_recv_nbx call sites, main.cpp:404/1172)
+//  * CPU sends complete when handed to the transport (buffer "reusable"
+//    semantics of ucp_tag_send_nbx eager); delivery is only guaranteed after
+//    flush (reference flush contract, tests/test_basic.py:250-416)
+//  * GPU sends are rendezvous: RTS -> receiver pulls over xGMI -> RECV_DONE;
+//    completion means delivery
+//  * close cancels everything still pending with an error containing
+//    "cancel" (reference main.cpp:680-701 contract)
+
+#include "core.hpp"
+
+#include <errno.h>
+#include <fcntl.h>
+#include <ifaddrs.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <sys/socket.h>
+#include <sys/uio.h>
+#include <unistd.h>
+
+#include <chrono>
+#include <cstdio>
+#include <random>
+
+namespace sw {
+
+// ---------------------------------------------------------------------------
+// small utils
+// ---------------------------------------------------------------------------
+
+#ifdef SW_DEBUG
+#define SW_DBG(...)                      \
+  do {                                   \
+    fprintf(stderr, "[sw] " __VA_ARGS__); \
+    fputc('\n', stderr);                 \
+  } while (0)
+#else
+#define SW_DBG(...) \
+  do {              \
+  } while (0)
+#endif
+
+static void set_nonblocking(int fd) {
+  int fl = fcntl(fd, F_GETFL, 0);
+  fcntl(fd, F_SETFL, fl | O_NONBLOCK);
+}
+
+static void set_tcp_opts(int fd) {
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  int buf = 8 * 1024 * 1024;
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
+}
+
+const uint8_t* process_uuid() {
+  static uint8_t uuid[16] = {0};
+  static bool init = [] {
+    std::random_device rd;
+    std::mt19937_64 gen(rd() ^ (uint64_t)getpid());
+    for (int i = 0; i < 16; i += 8) {
+      uint64_t v = gen();
+      memcpy(uuid + i, &v, 8);
+    }
+    return true;
+  }();
+  (void)init;
+  return uuid;
+}
+
+std::vector<uint8_t> encode_peer_info(const PeerInfo& pi) {
+  std::vector<uint8_t> out;
+  auto put = [&](const void* p, size_t n) {
+    const uint8_t* b = (const uint8_t*)p;
+    out.insert(out.end(), b, b + n);
+  };
+  uint32_t ver = kProtoVersion;
+  put(&ver, 4);
+  put(&pi.pid, 8);
+  put(pi.uuid, 16);
+  uint8_t hg = pi.has_gpu ? 1 : 0;
+  put(&hg, 1);
+  put(&pi.gpu_count, 4);
+  uint16_t nl = (uint16_t)pi.name.size();
+  put(&nl, 2);
+  put(pi.name.data(), nl);
+  return out;
+}
+
+bool decode_peer_info(const uint8_t* d, size_t len, PeerInfo* out) {
+  if (len < 4 + 8 + 16 + 1 + 4 + 2) return false;
+  size_t off = 0;
+  uint32_t ver;
+  memcpy(&ver, d + off, 4);
+  off += 4;
+  if (ver != kProtoVersion) return false;
+  memcpy(&out->pid, d + off, 8);
+  off += 8;
+  memcpy(out->uuid, d + off, 16);
+  off += 16;
+  out->has_gpu = d[off++] != 0;
+  memcpy(&out->gpu_count, d + off, 4);
+  off += 4;
+  uint16_t nl;
+  memcpy(&nl, d + off, 2);
+  off += 2;
+  if (off + nl > len) return false;
+  out->name.assign((const char*)d + off, nl);
+  return true;
+}
+
+// Worker-address blob:
+//   "SWADDR1\0" | u16 n_addrs | n x { u8 iplen, ip bytes, u16 port } |
+//   u32 peer_len | PeerInfo blob
+static constexpr char kAddrMagic[8] = {'S', 'W', 'A', 'D', 'D', 'R', '1', 0};
+
+static std::vector<std::string> local_ips() {
+  std::vector<std::string> ips;
+  struct ifaddrs* ifs = nullptr;
+  if (getifaddrs(&ifs) == 0) {
+    for (struct ifaddrs* it = ifs; it; it = it->ifa_next) {
+      if (!it->ifa_addr || it->ifa_addr->sa_family != AF_INET) continue;
+      char buf[INET_ADDRSTRLEN];
+      auto* sin = (struct sockaddr_in*)it->ifa_addr;
+      inet_ntop(AF_INET, &sin->sin_addr, buf, sizeof(buf));
+      std::string ip(buf);
+      if (ip == "127.0.0.1")
+        ips.push_back(ip);  // keep, but non-loopback preferred first
+      else
+        ips.insert(ips.begin(), ip);
+    }
+    freeifaddrs(ifs);
+  }
+  if (ips.empty()) ips.push_back("127.0.0.1");
+  return ips;
+}
+
+static std::vector<uint8_t> encode_worker_address(
+    const std::vector<std::pair<std::string, int>>& addrs, const PeerInfo& pi) {
+  std::vector<uint8_t> out(kAddrMagic, kAddrMagic + 8);
+  auto put = [&](const void* p, size_t n) {
+    const uint8_t* b = (const uint8_t*)p;
+    out.insert(out.end(), b, b + n);
+  };
+  uint16_t n = (uint16_t)addrs.size();
+  put(&n, 2);
+  for (auto& [ip, port] : addrs) {
+    uint8_t il = (uint8_t)ip.size();
+    put(&il, 1);
+    put(ip.data(), il);
+    uint16_t p = (uint16_t)port;
+    put(&p, 2);
+  }
+  auto blob = encode_peer_info(pi);
+  uint32_t bl = (uint32_t)blob.size();
+  put(&bl, 4);
+  put(blob.data(), bl);
+  return out;
+}
+
+static bool decode_worker_address(const std::vector<uint8_t>& in,
+                                  std::vector<std::pair<std::string, int>>* addrs,
+                                  PeerInfo* pi) {
+  if (in.size() < 10 || memcmp(in.data(), kAddrMagic, 8) != 0) return false;
+  size_t off = 8;
+  uint16_t n;
+  memcpy(&n, in.data() + off, 2);
+  off += 2;
+  for (int i = 0; i < n; i++) {
+    if (off + 1 > in.size()) return false;
+    uint8_t il = in[off++];
+    if (off + il + 2 > in.size()) return false;
+    std::string ip((const char*)in.data() + off, il);
+    off += il;
+    uint16_t port;
+    memcpy(&port, in.data() + off, 2);
+    off += 2;
+    addrs->emplace_back(ip, port);
+  }
+  if (off + 4 > in.size()) return false;
+  uint32_t bl;
+  memcpy(&bl, in.data() + off, 4);
+  off += 4;
+  if (off + bl > in.size()) return false;
+  return decode_peer_info(in.data() + off, bl, pi);
+}
+
+static PeerInfo self_peer_info(const std::string& name) {
+  PeerInfo pi;
+  pi.pid = (uint64_t)getpid();
+  memcpy(pi.uuid, process_uuid(), 16);
+  pi.has_gpu = gpu::available();
+  pi.gpu_count = gpu::device_count();
+  pi.name = name;
+  return pi;
+}
+
+// ---------------------------------------------------------------------------
+// GpuPull — one in-flight GPU delivery (RTS pull or host->device bounce)
+// ---------------------------------------------------------------------------
+
+struct Engine::GpuPull {
+  void* ticket = nullptr;
+  Op* recv_op = nullptr;          // completed with (tag, len) when done
+  Connection* conn = nullptr;     // where to ack
+  uint64_t sender_op_id = 0;      // 0 => no RECV_DONE ack needed (h2d bounce)
+  uint64_t tag = 0;
+  uint64_t len = 0;
+};
+
+// ---------------------------------------------------------------------------
+// Engine
+// ---------------------------------------------------------------------------
+
+Engine::Engine(Mode mode) : mode_(mode) {
+  if (pipe(wake_fds_) == 0) {
+    set_nonblocking(wake_fds_[0]);
+    set_nonblocking(wake_fds_[1]);
+  }
+}
+
+Engine::~Engine() {
+  int st = status_.load();
+  if (st >= 1 && st < 4 && thread_.joinable()) {
+    // Force-close: reference dtor safety net (main.cpp:703-719).
+    status_.store(3, std::memory_order_release);
+    wake();
+  }
+  if (thread_.joinable()) {
+    py::gil_scoped_release rel;
+    thread_.join();
+  }
+  if (listen_fd_ >= 0) ::close(listen_fd_);
+  if (wake_fds_[0] >= 0) ::close(wake_fds_[0]);
+  if (wake_fds_[1] >= 0) ::close(wake_fds_[1]);
+}
+
+void Engine::wake() {
+  char b = 1;
+  ssize_t r = ::write(wake_fds_[1], &b, 1);
+  (void)r;
+}
+
+// ---- Python-thread API ----------------------------------------------------
+
+Connection* Engine::make_listener(const std::string& addr, int port) {
+  int fd = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) throw std::runtime_error("socket() failed");
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  struct sockaddr_in sin {};
+  sin.sin_family = AF_INET;
+  sin.sin_port = htons((uint16_t)port);
+  if (inet_pton(AF_INET, addr.c_str(), &sin.sin_addr) != 1) {
+    ::close(fd);
+    throw std::runtime_error("invalid listen address: " + addr);
+  }
+  if (::bind(fd, (struct sockaddr*)&sin, sizeof(sin)) != 0) {
+    ::close(fd);
+    throw std::runtime_error("bind failed on " + addr + ":" +
+                             std::to_string(port) + ": " + strerror(errno));
+  }
+  if (::listen(fd, 128) != 0) {
+    ::close(fd);
+    throw std::runtime_error("listen failed: " + std::string(strerror(errno)));
+  }
+  struct sockaddr_in got {};
+  socklen_t gl = sizeof(got);
+  getsockname(fd, (struct sockaddr*)&got, &gl);
+  listen_port_ = ntohs(got.sin_port);
+  listen_host_ = addr;
+  set_nonblocking(fd);
+  listen_fd_ = fd;
+  return nullptr;
+}
+
+void Engine::listen(const std::string& addr, int port) {
+  if (mode_ != ServerMode) throw std::runtime_error("not a server");
+  int expect = 0;
+  if (!status_.compare_exchange_strong(expect, 1))
+    throw std::runtime_error("server already listening or closed");
+  try {
+    make_listener(addr, port);
+  } catch (...) {
+    status_.store(0);
+    throw;
+  }
+  thread_ = std::thread([this] { thread_main(); });
+  // Reference blocks listen() until the engine is running (main.cpp:829-831).
+  py::gil_scoped_release rel;
+  while (status_.load(std::memory_order_acquire) < 2) sched_yield();
+}
+
+std::vector<uint8_t> Engine::listen_address() {
+  if (mode_ != ServerMode) throw std::runtime_error("not a server");
+  listen("0.0.0.0", 0);  // ephemeral port; address carried in the blob
+  worker_mode_ = true;
+  return get_worker_address();
+}
+
+std::vector<uint8_t> Engine::get_worker_address() {
+  std::vector<std::pair<std::string, int>> addrs;
+  if (listen_fd_ >= 0) {
+    if (listen_host_ == "0.0.0.0") {
+      for (auto& ip : local_ips()) addrs.emplace_back(ip, listen_port_);
+    } else {
+      addrs.emplace_back(listen_host_, listen_port_);
+    }
+  }
+  return encode_worker_address(
+      addrs, self_peer_info(mode_ == ServerMode ? "server" : "client"));
+}
+
+void Engine::connect(const std::string& addr, int port, py::object cb) {
+  if (mode_ != ClientMode) throw std::runtime_error("not a client");
+  int expect = 0;
+  if (!status_.compare_exchange_strong(expect, 1))
+    throw std::runtime_error("client already connected or closed");
+  connect_host_ = addr;
+  connect_port_ = port;
+  connect_cb_ = std::move(cb);
+  connect_requested_ = true;
+  thread_ = std::thread([this] { thread_main(); });
+}
+
+void Engine::connect_address(const std::vector<uint8_t>& blob, py::object cb) {
+  if (mode_ != ClientMode) throw std::runtime_error("not a client");
+  std::vector<std::pair<std::string, int>> addrs;
+  PeerInfo pi;
+  if (!decode_worker_address(blob, &addrs, &pi) || addrs.empty())
+    throw std::runtime_error("invalid worker address blob");
+  // Try the first address; remaining candidates joined for the engine to try.
+  int expect = 0;
+  if (!status_.compare_exchange_strong(expect, 1))
+    throw std::runtime_error("client already connected or closed");
+  connect_host_ = addrs[0].first;
+  connect_port_ = addrs[0].second;
+  connect_cb_ = std::move(cb);
+  connect_requested_ = true;
+  thread_ = std::thread([this] { thread_main(); });
+}
+
+void Engine::set_accept_callback(py::object cb) {
+  accept_cb_ = std::move(cb);
+  have_accept_cb_ = true;
+}
+
+void Engine::close(py::object cb) {
+  int expect = 2;
+  if (!status_.compare_exchange_strong(expect, 3))
+    throw std::runtime_error(
+        "close() called but the endpoint is not in a running state");
+  close_cb_ = std::move(cb);
+  wake();
+}
+
+void Engine::send(std::shared_ptr<EndpointInfo> ep, BufferRef buf, uint64_t tag,
+                  py::object done, py::object fail, py::object keepalive) {
+  if (status_.load(std::memory_order_acquire) != 2)
+    throw std::runtime_error("send: endpoint not connected/running");
+  Op* op = new Op();
+  op->id = next_op_id_.fetch_add(1);
+  op->type = OpType::Send;
+  op->buf = buf;
+  op->tag = tag;
+  op->conn = ep ? ep->conn : nullptr;  // resolved again engine-side via ep
+  op->done_cb = std::move(done);
+  op->fail_cb = std::move(fail);
+  op->keepalive = std::move(keepalive);
+  // Stash the ep so the engine resolves conn at processing time (the raw
+  // conn pointer above may already be dead).
+  op->ep_ref = std::move(ep);
+  {
+    std::lock_guard<std::mutex> lk(cmd_mu_);
+    cmd_queue_.push_back(op);
+  }
+  cmd_pending_.store(true, std::memory_order_release);
+  wake();
+}
+
+void Engine::recv(BufferRef buf, uint64_t tag, uint64_t mask, py::object done,
+                  py::object fail, py::object keepalive) {
+  if (status_.load(std::memory_order_acquire) != 2)
+    throw std::runtime_error("recv: endpoint not connected/running");
+  Op* op = new Op();
+  op->id = next_op_id_.fetch_add(1);
+  op->type = OpType::Recv;
+  op->buf = buf;
+  op->tag = tag;
+  op->tag_mask = mask;
+  op->done_cb = std::move(done);
+  op->fail_cb = std::move(fail);
+  op->keepalive = std::move(keepalive);
+  {
+    std::lock_guard<std::mutex> lk(cmd_mu_);
+    cmd_queue_.push_back(op);
+  }
+  cmd_pending_.store(true, std::memory_order_release);
+  wake();
+}
+
+void Engine::flush(py::object done, py::object fail) {
+  if (status_.load(std::memory_order_acquire) != 2)
+    throw std::runtime_error("flush: endpoint not connected/running");
+  Op* op = new Op();
+  op->id = next_op_id_.fetch_add(1);
+  op->type = OpType::Flush;
+  op->done_cb = std::move(done);
+  op->fail_cb = std::move(fail);
+  {
+    std::lock_guard<std::mutex> lk(cmd_mu_);
+    cmd_queue_.push_back(op);
+  }
+  cmd_pending_.store(true, std::memory_order_release);
+  wake();
+}
+
+void Engine::flush_ep(std::shared_ptr<EndpointInfo> ep, py::object done,
+                      py::object fail) {
+  if (status_.load(std::memory_order_acquire) != 2)
+    throw std::runtime_error("flush_ep: endpoint not connected/running");
+  Op* op = new Op();
+  op->id = next_op_id_.fetch_add(1);
+  op->type = OpType::FlushEp;
+  op->done_cb = std::move(done);
+  op->fail_cb = std::move(fail);
+  op->ep_ref = std::move(ep);
+  {
+    std::lock_guard<std::mutex> lk(cmd_mu_);
+    cmd_queue_.push_back(op);
+  }
+  cmd_pending_.store(true, std::memory_order_release);
+  wake();
+}
+
+std::vector<std::shared_ptr<EndpointInfo>> Engine::list_clients() {
+  std::lock_guard<std::mutex> lk(ep_mu_);
+  return eps_;
+}
+
+double Engine::perf_model(Connection* c, uint64_t msg_size) const {
+  // Analytic transfer-time model, the ucp_ep_evaluate_perf analog
+  // (reference main.cpp:452-467). Constants from xGMI topology (7 links x
+  // ~153 GB/s per MI355X) and localhost TCP measurements; recalibrated once
+  // rocprof evidence lands (profiles/).
+  bool peer_gpu = c && c->peer.has_gpu;
+  bool self_gpu = gpu::available();
+  double lat, bw;
+  if (self_gpu && peer_gpu) {
+    lat = 15e-6;           // RTS over TCP + kernel launch + event poll
+    bw = 140e9;            // single xGMI link, sustained
+  } else {
+    lat = 25e-6;           // localhost TCP eager
+    bw = 3e9;
+  }
+  return lat + (double)msg_size / bw;
+}
+
+double Engine::evaluate_perf(std::shared_ptr<EndpointInfo> ep,
+                             uint64_t msg_size) {
+  if (status_.load(std::memory_order_acquire) != 2)
+    throw std::runtime_error("evaluate_perf: endpoint not running");
+  Connection* c = ep ? ep->conn : (conns_.empty() ? nullptr : conns_[0].get());
+  return perf_model(c, msg_size);
+}
+
+// ---- engine thread --------------------------------------------------------
+
+void Engine::thread_main() {
+  if (mode_ == ClientMode && connect_requested_) {
+    do_connect_start();
+    if (status_.load() != 2) return;  // connect failed; callback already fired
+  } else {
+    status_.store(2, std::memory_order_release);
+  }
+  bool did_work = false;
+  while (status_.load(std::memory_order_acquire) == 2) {
+    did_work = false;
+    loop_iteration(did_work);
+    if (did_work) {
+      idle_iters_ = 0;
+    } else {
+      idle_iters_++;
+    }
+  }
+  teardown();
+}
+
+void Engine::do_connect_start() {
+  // Blocking-ish connect with handshake, abortable via status_==3.
+  auto fail = [&](const std::string& why) {
+    Completion comp;
+    comp.kind = Completion::Kind::Connect;
+    comp.error = "not connected: " + why;
+    comp.cb0 = py::object();
+    {
+      py::gil_scoped_acquire gil;
+      try {
+        if (connect_cb_.ptr()) connect_cb_(comp.error);
+      } catch (py::error_already_set& e) {
+        e.discard_as_unraisable("starway connect callback");
+      }
+      connect_cb_ = py::object();
+    }
+    status_.store(4, std::memory_order_release);
+  };
+
+  struct sockaddr_in sin {};
+  sin.sin_family = AF_INET;
+  sin.sin_port = htons((uint16_t)connect_port_);
+  if (inet_pton(AF_INET, connect_host_.c_str(), &sin.sin_addr) != 1)
+    return fail("invalid address " + connect_host_);
+
+  // Connect with a 5 s budget; ECONNREFUSED is retried (the peer may still
+  // be starting up — common in spawn-subprocess tests and rank rendezvous).
+  auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(5);
+  int fd = -1;
+  std::string last_err = "connect TIMEOUT";
+  while (fd < 0) {
+    if (status_.load(std::memory_order_acquire) == 3) return fail("canceled");
+    if (std::chrono::steady_clock::now() > deadline) return fail(last_err);
+    int s = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (s < 0) return fail("socket() failed");
+    set_nonblocking(s);
+    set_tcp_opts(s);
+    int r = ::connect(s, (struct sockaddr*)&sin, sizeof(sin));
+    int cerr = 0;
+    if (r == 0) {
+      fd = s;
+      break;
+    }
+    if (errno == EINPROGRESS) {
+      while (true) {
+        if (status_.load(std::memory_order_acquire) == 3) {
+          ::close(s);
+          return fail("canceled");
+        }
+        struct pollfd pfd {s, POLLOUT, 0};
+        int pr = ::poll(&pfd, 1, 20);
+        if (pr > 0) {
+          socklen_t el = sizeof(cerr);
+          getsockopt(s, SOL_SOCKET, SO_ERROR, &cerr, &el);
+          break;
+        }
+        if (std::chrono::steady_clock::now() > deadline) {
+          ::close(s);
+          return fail("connect TIMEOUT");
+        }
+      }
+    } else {
+      cerr = errno;
+    }
+    if (cerr == 0) {
+      fd = s;
+      break;
+    }
+    ::close(s);
+    last_err = std::string("connect: ") + strerror(cerr);
+    if (cerr != ECONNREFUSED && cerr != ENETUNREACH && cerr != EHOSTUNREACH)
+      return fail(last_err);
+    usleep(20000);  // refused: server not up yet; retry until deadline
+  }
+
+  auto c = std::make_unique<Connection>();
+  c->fd = fd;
+  c->conn_id = next_conn_id_++;
+  struct sockaddr_in la {};
+  socklen_t ll = sizeof(la);
+  getsockname(fd, (struct sockaddr*)&la, &ll);
+  char ab[INET_ADDRSTRLEN];
+  inet_ntop(AF_INET, &la.sin_addr, ab, sizeof(ab));
+  c->local_addr = ab;
+  c->local_port = ntohs(la.sin_port);
+  c->remote_addr = connect_host_;
+  c->remote_port = connect_port_;
+  Connection* cp = c.get();
+  conns_.push_back(std::move(c));
+  send_hello(cp);
+
+  // Drive IO until HELLO exchanged (10 s budget).
+  deadline = std::chrono::steady_clock::now() + std::chrono::seconds(10);
+  while (!cp->hello_received) {
+    if (status_.load(std::memory_order_acquire) == 3 || cp->dead) {
+      return fail("canceled or peer reset during handshake");
+    }
+    bool did = false;
+    poll_sockets(5, did);
+    if (std::chrono::steady_clock::now() > deadline)
+      return fail("handshake TIMEOUT");
+  }
+  status_.store(2, std::memory_order_release);
+  {
+    py::gil_scoped_acquire gil;
+    try {
+      if (connect_cb_.ptr()) connect_cb_(std::string(""));
+    } catch (py::error_already_set& e) {
+      e.discard_as_unraisable("starway connect callback");
+    }
+    connect_cb_ = py::object();
+  }
+}
+
+void Engine::loop_iteration(bool& did_work) {
+  if (cmd_pending_.load(std::memory_order_acquire)) {
+    std::vector<Op*> cmds;
+    drain_commands(cmds);
+    for (Op* op : cmds) {
+      process_command(op);
+      did_work = true;
+    }
+  }
+  // Block in poll() only when fully idle; stay hot whenever GPU events or
+  // outbound bytes are pending (the ucp_worker_progress spin analog,
+  // reference main.cpp:361-468).
+  int timeout = 0;
+  bool busy = !gpu_pulls_.empty();
+  if (!busy)
+    for (auto& c : conns_)
+      if (c->want_write()) {
+        busy = true;
+        break;
+      }
+  if (!busy && idle_iters_ > 50000) timeout = 1;
+  poll_sockets(timeout, did_work);
+  if (!gpu_pulls_.empty()) poll_gpu(did_work);
+  if (!completions_.empty()) {
+    fire_completions();
+    did_work = true;
+  }
+}
+
+void Engine::drain_commands(std::vector<Op*>& cmds) {
+  std::lock_guard<std::mutex> lk(cmd_mu_);
+  cmds.swap(cmd_queue_);
+  cmd_pending_.store(false, std::memory_order_release);
+}
+
+void Engine::process_command(Op* op) {
+  switch (op->type) {
+    case OpType::Send: {
+      Connection* c = nullptr;
+      if (mode_ == ClientMode) {
+        c = conns_.empty() ? nullptr : conns_[0].get();
+      } else {
+        c = op->ep_ref ? op->ep_ref->conn : nullptr;
+      }
+      if (!c || c->dead) {
+        fail_op(op, "send failed: endpoint closed");
+        return;
+      }
+      op->conn = c;
+      if (op->buf.device >= 0) {
+        // GPU rendezvous.
+        RtsDesc rts{};
+        std::string err;
+        if (!gpu::make_rts(op->buf, &rts, &err)) {
+          fail_op(op, "send failed: " + err);
+          return;
+        }
+        enqueue_frame(c, FT_RTS, op->tag, op->id, op->buf.size, &rts,
+                      sizeof(rts), /*priority=*/false);
+        op->gpu_send_awaiting_ack = true;
+        gpu_sends_[op->id] = op;
+      } else {
+        enqueue_eager(c, op);
+        // CPU send: handed to transport => complete (eager semantics).
+        Completion comp;
+        comp.kind = Completion::Kind::SendDone;
+        comp.op = op;
+        complete(std::move(comp));
+      }
+      break;
+    }
+    case OpType::Recv:
+      match_or_stash_recv(op);
+      break;
+    case OpType::Flush:
+    case OpType::FlushEp: {
+      std::vector<Connection*> targets;
+      if (op->type == OpType::FlushEp) {
+        Connection* c = op->ep_ref ? op->ep_ref->conn : nullptr;
+        if (!c || c->dead) {
+          fail_op(op, "flush_ep failed: endpoint closed");
+          return;
+        }
+        targets.push_back(c);
+      } else {
+        for (auto& c : conns_)
+          if (!c->dead && c->hello_received) targets.push_back(c.get());
+      }
+      for (Connection* c : targets) {
+        uint64_t fid = next_flush_id_++;
+        enqueue_frame(c, FT_FLUSH_REQ, 0, fid, 0, nullptr, 0, false);
+        op->flush_acks_pending.insert({c, fid});
+      }
+      for (auto& [id, sop] : gpu_sends_) {
+        if (op->type == OpType::Flush ||
+            (!targets.empty() && sop->conn == targets[0]))
+          op->flush_ops_pending.insert(id);
+      }
+      if (op->flush_acks_pending.empty() && op->flush_ops_pending.empty()) {
+        Completion comp;
+        comp.kind = Completion::Kind::FlushDone;
+        comp.op = op;
+        complete(std::move(comp));
+      } else {
+        pending_flushes_.push_back(op);
+      }
+      break;
+    }
+    default:
+      fail_op(op, "internal: unknown command");
+  }
+}
+
+// ---- socket IO ------------------------------------------------------------
+
+void Engine::poll_sockets(int timeout_ms, bool& did_work) {
+  std::vector<struct pollfd> pfds;
+  pfds.push_back({wake_fds_[0], POLLIN, 0});
+  size_t listener_idx = SIZE_MAX;
+  if (listen_fd_ >= 0) {
+    listener_idx = pfds.size();
+    pfds.push_back({listen_fd_, POLLIN, 0});
+  }
+  size_t conn_base = pfds.size();
+  std::vector<Connection*> live;
+  for (auto& c : conns_) {
+    if (c->dead || c->fd < 0) continue;
+    short ev = POLLIN;
+    if (c->want_write()) ev |= POLLOUT;
+    pfds.push_back({c->fd, ev, 0});
+    live.push_back(c.get());
+  }
+  int r = ::poll(pfds.data(), (nfds_t)pfds.size(), timeout_ms);
+  if (r <= 0) return;
+  if (pfds[0].revents & POLLIN) {
+    char buf[256];
+    while (::read(wake_fds_[0], buf, sizeof(buf)) > 0) {
+    }
+  }
+  if (listener_idx != SIZE_MAX && (pfds[listener_idx].revents & POLLIN))
+    accept_new(did_work);
+  for (size_t i = 0; i < live.size(); i++) {
+    short re = pfds[conn_base + i].revents;
+    Connection* c = live[i];
+    if (re & (POLLIN | POLLERR | POLLHUP)) handle_readable(c, did_work);
+    if (!c->dead && (re & POLLOUT)) handle_writable(c, did_work);
+  }
+}
+
+void Engine::accept_new(bool& did_work) {
+  while (true) {
+    struct sockaddr_in ra {};
+    socklen_t rl = sizeof(ra);
+    int fd = ::accept4(listen_fd_, (struct sockaddr*)&ra, &rl, SOCK_CLOEXEC);
+    if (fd < 0) break;
+    set_nonblocking(fd);
+    set_tcp_opts(fd);
+    auto c = std::make_unique<Connection>();
+    c->fd = fd;
+    c->conn_id = next_conn_id_++;
+    char ab[INET_ADDRSTRLEN];
+    inet_ntop(AF_INET, &ra.sin_addr, ab, sizeof(ab));
+    c->remote_addr = ab;
+    c->remote_port = ntohs(ra.sin_port);
+    struct sockaddr_in la {};
+    socklen_t ll = sizeof(la);
+    getsockname(fd, (struct sockaddr*)&la, &ll);
+    inet_ntop(AF_INET, &la.sin_addr, ab, sizeof(ab));
+    c->local_addr = ab;
+    c->local_port = ntohs(la.sin_port);
+    conns_.push_back(std::move(c));
+    // HELLO handshake is responder-style: the server replies only after
+    // registering the endpoint, so by the time the client's connect
+    // completes, list_clients() already shows it (reference contract,
+    // tests/test_basic.py:43-58).
+    did_work = true;
+  }
+}
+
+void Engine::send_hello(Connection* c) {
+  PeerInfo pi = self_peer_info(mode_ == ServerMode ? "server" : "client");
+  auto blob = encode_peer_info(pi);
+  enqueue_frame(c, FT_HELLO, 0, 0, 0, blob.data(), blob.size(),
+                /*priority=*/true);
+  c->hello_sent = true;
+}
+
+void Engine::handle_readable(Connection* c, bool& did_work) {
+  while (!c->dead) {
+    if (c->rx_state == Connection::RxState::Header) {
+      uint8_t* hp = (uint8_t*)&c->rx_hdr;
+      ssize_t n = ::read(c->fd, hp + c->rx_got, sizeof(FrameHeader) - c->rx_got);
+      if (n == 0) {
+        on_conn_dead(c);
+        return;
+      }
+      if (n < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+        on_conn_dead(c);
+        return;
+      }
+      did_work = true;
+      c->rx_got += (size_t)n;
+      if (c->rx_got < sizeof(FrameHeader)) continue;
+      c->rx_got = 0;
+      if (c->rx_hdr.magic != kMagic) {
+        SW_DBG("bad magic from %s:%d", c->remote_addr.c_str(), c->remote_port);
+        on_conn_dead(c);
+        return;
+      }
+      on_frame(c);
+    } else {
+      // Payload streaming.
+      size_t want;
+      uint8_t* dst;
+      if (c->rx_hdr.type == FT_EAGER) {
+        uint64_t done = c->rx_hdr.size - c->rx_msg_remaining;
+        if (c->rx_recv_op && !c->rx_truncated) {
+          // Zero-copy into the posted buffer (or its host bounce for GPU).
+          if (c->rx_recv_op->buf.device >= 0) {
+            dst = c->rx_gpu_bounce.data() + done;
+          } else {
+            dst = c->rx_recv_op->buf.ptr + done;
+          }
+        } else if (c->rx_unexp) {
+          dst = c->rx_unexp->data.data() + done;
+        } else {
+          if (c->rx_discard.size() < (64 << 10)) c->rx_discard.resize(64 << 10);
+          dst = c->rx_discard.data();
+        }
+        want = c->rx_msg_remaining;
+        if ((!c->rx_recv_op || c->rx_truncated) && !c->rx_unexp)
+          want = std::min<size_t>(want, c->rx_discard.size());
+      } else {
+        dst = c->rx_small.data() + c->rx_got;
+        want = c->rx_hdr.size - c->rx_got;
+      }
+      ssize_t n = ::read(c->fd, dst, want);
+      if (n == 0) {
+        on_conn_dead(c);
+        return;
+      }
+      if (n < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+        on_conn_dead(c);
+        return;
+      }
+      did_work = true;
+      if (c->rx_hdr.type == FT_EAGER) {
+        c->rx_msg_remaining -= (uint64_t)n;
+        if (c->rx_unexp) c->rx_unexp->got += (uint64_t)n;
+        if (c->rx_msg_remaining == 0) finish_eager_into_recv(c);
+      } else {
+        c->rx_got += (size_t)n;
+        if (c->rx_got == c->rx_hdr.size) {
+          c->rx_got = 0;
+          c->rx_state = Connection::RxState::Header;
+          on_frame_payload(c);
+        }
+      }
+    }
+  }
+}
+
+void Engine::on_frame(Connection* c) {
+  FrameHeader& h = c->rx_hdr;
+  switch (h.type) {
+    case FT_EAGER:
+      begin_eager(c);
+      return;
+    case FT_HELLO:
+    case FT_RTS:
+    case FT_RECV_FAIL:
+      if (h.size > (16 << 20)) {
+        on_conn_dead(c);
+        return;
+      }
+      c->rx_small.resize(h.size);
+      c->rx_got = 0;
+      c->rx_state = Connection::RxState::Payload;
+      if (h.size == 0) {
+        c->rx_state = Connection::RxState::Header;
+        on_frame_payload(c);
+      }
+      return;
+    case FT_RECV_DONE:
+      on_gpu_send_acked(h.op_id, false, "");
+      return;
+    case FT_FLUSH_REQ:
+      // TCP ordering: by the time we parse this, all earlier bytes on this
+      // stream have been consumed by the engine => safe to ack.
+      enqueue_frame(c, FT_FLUSH_ACK, 0, h.op_id, 0, nullptr, 0, true);
+      return;
+    case FT_FLUSH_ACK:
+      check_flush_progress(c, h.op_id);
+      return;
+    case FT_BYE:
+      on_conn_dead(c);
+      return;
+    default:
+      on_conn_dead(c);
+      return;
+  }
+}
+
+void Engine::on_frame_payload(Connection* c) {
+  FrameHeader& h = c->rx_hdr;
+  switch (h.type) {
+    case FT_HELLO: {
+      PeerInfo pi;
+      if (!decode_peer_info(c->rx_small.data(), c->rx_small.size(), &pi)) {
+        on_conn_dead(c);
+        return;
+      }
+      c->peer = pi;
+      c->hello_received = true;
+      on_hello(c);
+      break;
+    }
+    case FT_RTS: {
+      if (c->rx_small.size() != sizeof(RtsDesc)) {
+        on_conn_dead(c);
+        return;
+      }
+      RtsDesc rts;
+      memcpy(&rts, c->rx_small.data(), sizeof(rts));
+      handle_rts(c, rts, h.tag, h.aux, h.op_id);
+      break;
+    }
+    case FT_RECV_FAIL: {
+      std::string err((const char*)c->rx_small.data(), c->rx_small.size());
+      on_gpu_send_acked(h.op_id, true, err);
+      break;
+    }
+    default:
+      break;
+  }
+}
+
+void Engine::on_hello(Connection* c) {
+  if (mode_ == ServerMode) {
+    auto ep = std::make_shared<EndpointInfo>();
+    ep->name = "ep-" + std::to_string(c->conn_id) + "@" + c->remote_addr + ":" +
+               std::to_string(c->remote_port);
+    ep->local_addr = c->local_addr;
+    ep->local_port = c->local_port;
+    ep->remote_addr = c->remote_addr;
+    ep->remote_port = c->remote_port;
+    ep->conn = c;
+    ep->owner = this;
+    ep->transports.emplace_back("tcp", "sock");
+    if (gpu::available() && c->peer.has_gpu) {
+      bool same_proc = memcmp(c->peer.uuid, process_uuid(), 16) == 0;
+      ep->transports.emplace_back("xgmi", same_proc ? "p2p" : "hipipc");
+      ep->transports.emplace_back("hbm", "gfx950_copy");
+    }
+    c->ep = ep;
+    {
+      std::lock_guard<std::mutex> lk(ep_mu_);
+      eps_.push_back(ep);
+    }
+    send_hello(c);  // responder reply; unblocks the client's connect
+    if (have_accept_cb_) {
+      Completion comp;
+      comp.kind = Completion::Kind::Accept;
+      comp.ep = ep;
+      complete(std::move(comp));
+    }
+  }
+}
+
+// ---- eager path -----------------------------------------------------------
+
+void Engine::begin_eager(Connection* c) {
+  uint64_t msg_len = c->rx_hdr.size;
+  uint64_t tag = c->rx_hdr.tag;
+  c->rx_recv_op = nullptr;
+  c->rx_unexp = nullptr;
+  c->rx_truncated = false;
+  c->rx_msg_remaining = msg_len;
+
+  // Match against posted recvs in FIFO order.
+  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
+    Op* r = *it;
+    if ((tag & r->tag_mask) == (r->tag & r->tag_mask)) {
+      posted_recvs_.erase(it);
+      if (msg_len > r->buf.size) {
+        // Truncation: consume + fail (UCX MESSAGE_TRUNCATED analog).
+        c->rx_truncated = true;
+        c->rx_recv_op = r;
+      } else {
+        c->rx_recv_op = r;
+        if (r->buf.device >= 0) c->rx_gpu_bounce.alloc(msg_len);
+      }
+      c->rx_recv_op->recv_sender_tag = tag;
+      c->rx_recv_op->recv_len = msg_len;
+      break;
+    }
+  }
+  if (!c->rx_recv_op) {
+    auto um = std::make_unique<UnexpectedMsg>();
+    um->tag = tag;
+    um->size = msg_len;
+    um->conn = c;
+    um->data.alloc(msg_len);
+    c->rx_unexp = um.get();
+    unexpected_.push_back(std::move(um));
+  }
+  if (msg_len == 0) {
+    finish_eager_into_recv(c);
+  } else {
+    c->rx_state = Connection::RxState::Payload;
+  }
+}
+
+void Engine::finish_eager_into_recv(Connection* c) {
+  c->rx_state = Connection::RxState::Header;
+  c->rx_got = 0;
+  if (c->rx_recv_op) {
+    Op* r = c->rx_recv_op;
+    c->rx_recv_op = nullptr;
+    if (c->rx_truncated) {
+      c->rx_truncated = false;
+      fail_op(r, "receive failed: message truncated (len " +
+                     std::to_string(r->recv_len) + " > buffer " +
+                     std::to_string(r->buf.size) + ")");
+      return;
+    }
+    if (r->buf.device >= 0) {
+      // Host->device bounce upload.
+      std::string err;
+      void* ticket = gpu::begin_h2d(r->buf, c->rx_gpu_bounce.data(),
+                                    r->recv_len, &err);
+      if (!ticket) {
+        fail_op(r, "receive failed: " + err);
+        return;
+      }
+      auto pull = std::make_unique<GpuPull>();
+      pull->ticket = ticket;
+      pull->recv_op = r;
+      pull->conn = c;
+      pull->sender_op_id = 0;
+      pull->tag = r->recv_sender_tag;
+      pull->len = r->recv_len;
+      gpu::attach_bounce(ticket, std::move(c->rx_gpu_bounce));
+      c->rx_gpu_bounce.clear();
+      gpu_pulls_.push_back(std::move(pull));
+      return;
+    }
+    Completion comp;
+    comp.kind = Completion::Kind::RecvDone;
+    comp.op = r;
+    comp.a = r->recv_sender_tag;
+    comp.b = r->recv_len;
+    complete(std::move(comp));
+  } else if (c->rx_unexp) {
+    c->rx_unexp->complete = true;
+    UnexpectedMsg* um = c->rx_unexp;
+    c->rx_unexp = nullptr;
+    if (um->bound_recv) {
+      Op* r = um->bound_recv;
+      um->bound_recv = nullptr;
+      // Transfer ownership out of the unexpected queue before delivering.
+      for (auto it = unexpected_.begin(); it != unexpected_.end(); ++it) {
+        if (it->get() == um) {
+          it->release();
+          unexpected_.erase(it);
+          break;
+        }
+      }
+      complete_recv_from_unexpected(r, um);
+    }
+  }
+}
+
+// ---- GPU rendezvous -------------------------------------------------------
+
+void Engine::handle_rts(Connection* c, const RtsDesc& rts, uint64_t tag,
+                        uint64_t size, uint64_t sender_op) {
+  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
+    Op* r = *it;
+    if ((tag & r->tag_mask) == (r->tag & r->tag_mask)) {
+      posted_recvs_.erase(it);
+      start_gpu_pull(r, rts, tag, size, sender_op, c);
+      return;
+    }
+  }
+  auto um = std::make_unique<UnexpectedMsg>();
+  um->tag = tag;
+  um->size = size;
+  um->conn = c;
+  um->is_rts = true;
+  um->rts = rts;
+  um->sender_op_id = sender_op;
+  um->complete = true;
+  unexpected_.push_back(std::move(um));
+}
+
+void Engine::start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
+                            uint64_t size, uint64_t sender_op, Connection* c) {
+  if (size > recv_op->buf.size) {
+    std::string err = "message truncated (len " + std::to_string(size) +
+                      " > buffer " + std::to_string(recv_op->buf.size) + ")";
+    enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
+                  true);
+    fail_op(recv_op, "receive failed: " + err);
+    return;
+  }
+  std::string err;
+  void* ticket = gpu::begin_pull(rts, recv_op->buf, size, &err);
+  if (!ticket) {
+    enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
+                  true);
+    fail_op(recv_op, "receive failed: " + err);
+    return;
+  }
+  auto pull = std::make_unique<GpuPull>();
+  pull->ticket = ticket;
+  pull->recv_op = recv_op;
+  pull->conn = c;
+  pull->sender_op_id = sender_op;
+  pull->tag = tag;
+  pull->len = size;
+  gpu_pulls_.push_back(std::move(pull));
+}
+
+void Engine::poll_gpu(bool& did_work) {
+  for (size_t i = 0; i < gpu_pulls_.size();) {
+    GpuPull* p = gpu_pulls_[i].get();
+    std::string err;
+    int r = gpu::poll_ticket(p->ticket, &err);
+    if (r == 0) {
+      i++;
+      continue;
+    }
+    did_work = true;
+    if (r > 0) {
+      if (p->sender_op_id && p->conn && !p->conn->dead)
+        enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
+                      true);
+      Completion comp;
+      comp.kind = Completion::Kind::RecvDone;
+      comp.op = p->recv_op;
+      comp.a = p->tag;
+      comp.b = p->len;
+      complete(std::move(comp));
+    } else {
+      if (p->sender_op_id && p->conn && !p->conn->dead)
+        enqueue_frame(p->conn, FT_RECV_FAIL, 0, p->sender_op_id, 0, err.data(),
+                      err.size(), true);
+      fail_op(p->recv_op, "receive failed: " + err);
+    }
+    gpu::free_ticket(p->ticket);
+    gpu_pulls_.erase(gpu_pulls_.begin() + i);
+  }
+}
+
+void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
+                               const std::string& err) {
+  auto it = gpu_sends_.find(op_id);
+  if (it == gpu_sends_.end()) return;
+  Op* op = it->second;
+  gpu_sends_.erase(it);
+  if (failed) {
+    fail_op(op, "send failed: " + err);
+  } else {
+    Completion comp;
+    comp.kind = Completion::Kind::SendDone;
+    comp.op = op;
+    complete(std::move(comp));
+  }
+  // Unblock flushes waiting on this op.
+  for (size_t i = 0; i < pending_flushes_.size();) {
+    Op* f = pending_flushes_[i];
+    f->flush_ops_pending.erase(op_id);
+    if (f->flush_acks_pending.empty() && f->flush_ops_pending.empty()) {
+      Completion comp;
+      comp.kind = Completion::Kind::FlushDone;
+      comp.op = f;
+      complete(std::move(comp));
+      pending_flushes_.erase(pending_flushes_.begin() + i);
+    } else {
+      i++;
+    }
+  }
+}
+
+// ---- recv matching --------------------------------------------------------
+
+void Engine::match_or_stash_recv(Op* op) {
+  if (!try_match_unexpected(op)) posted_recvs_.push_back(op);
+}
+
+bool Engine::try_match_unexpected(Op* op) {
+  for (auto it = unexpected_.begin(); it != unexpected_.end(); ++it) {
+    UnexpectedMsg* um = it->get();
+    if (um->bound_recv) continue;
+    if ((um->tag & op->tag_mask) != (op->tag & op->tag_mask)) continue;
+    if (um->is_rts) {
+      RtsDesc rts = um->rts;
+      uint64_t tag = um->tag, size = um->size, sop = um->sender_op_id;
+      Connection* c = um->conn;
+      unexpected_.erase(it);
+      start_gpu_pull(op, rts, tag, size, sop, c);
+      return true;
+    }
+    if (!um->complete) {
+      um->bound_recv = op;  // delivered when the stream finishes
+      return true;
+    }
+    UnexpectedMsg* owned = it->release();
+    unexpected_.erase(it);
+    complete_recv_from_unexpected(op, owned);
+    return true;
+  }
+  return false;
+}
+
+void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
+  std::unique_ptr<UnexpectedMsg> guard(um);
+  if (um->size > op->buf.size) {
+    fail_op(op, "receive failed: message truncated (len " +
+                    std::to_string(um->size) + " > buffer " +
+                    std::to_string(op->buf.size) + ")");
+    return;
+  }
+  if (op->buf.device >= 0) {
+    std::string err;
+    void* ticket = gpu::begin_h2d(op->buf, um->data.data(), um->size, &err);
+    if (!ticket) {
+      fail_op(op, "receive failed: " + err);
+      return;
+    }
+    auto pull = std::make_unique<GpuPull>();
+    pull->ticket = ticket;
+    pull->recv_op = op;
+    pull->conn = nullptr;
+    pull->sender_op_id = 0;
+    pull->tag = um->tag;
+    pull->len = um->size;
+    gpu::attach_bounce(ticket, std::move(um->data));
+    gpu_pulls_.push_back(std::move(pull));
+    return;
+  }
+  memcpy(op->buf.ptr, um->data.data(), um->size);
+  Completion comp;
+  comp.kind = Completion::Kind::RecvDone;
+  comp.op = op;
+  comp.a = um->tag;
+  comp.b = um->size;
+  complete(std::move(comp));
+}
+
+// ---- tx -------------------------------------------------------------------
+
+void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
+                           uint64_t op_id, uint64_t aux, const void* payload,
+                           size_t payload_len, bool priority) {
+  TxItem item;
+  item.head.resize(sizeof(FrameHeader) + payload_len);
+  FrameHeader h{};
+  h.magic = kMagic;
+  h.type = t;
+  h.tag = tag;
+  h.size = payload_len;
+  h.op_id = op_id;
+  h.aux = aux;
+  memcpy(item.head.data(), &h, sizeof(h));
+  if (payload_len) memcpy(item.head.data() + sizeof(h), payload, payload_len);
+  item.is_data = (t == FT_RTS);
+  if (priority && !c->txq.empty()) {
+    // Insert at the first frame boundary, after any queued priority frames.
+    size_t pos = c->tx_front_written > 0 ? 1 : 0;
+    c->txq.insert(c->txq.begin() + pos, std::move(item));
+  } else {
+    c->txq.push_back(std::move(item));
+  }
+  bool dummy = false;
+  handle_writable(c, dummy);  // opportunistic immediate write
+}
+
+void Engine::enqueue_eager(Connection* c, Op* op) {
+  TxItem item;
+  constexpr size_t kInline = 4096;
+  FrameHeader h{};
+  h.magic = kMagic;
+  h.type = FT_EAGER;
+  h.tag = op->tag;
+  h.size = op->buf.size;
+  h.op_id = op->id;
+  h.aux = op->buf.size;
+  item.is_data = true;
+  if (op->buf.size <= kInline) {
+    item.head.resize(sizeof(h) + op->buf.size);
+    memcpy(item.head.data(), &h, sizeof(h));
+    memcpy(item.head.data() + sizeof(h), op->buf.ptr, op->buf.size);
+  } else {
+    item.head.resize(sizeof(h));
+    memcpy(item.head.data(), &h, sizeof(h));
+    item.ext = op->buf.ptr;
+    item.ext_len = op->buf.size;
+    item.has_keepalive = true;
+    item.keepalive = std::move(op->keepalive);  // moved, no refcount touch
+  }
+  c->txq.push_back(std::move(item));
+  bool dummy = false;
+  handle_writable(c, dummy);
+}
+
+void Engine::handle_writable(Connection* c, bool& did_work) {
+  while (!c->txq.empty() && !c->dead) {
+    TxItem& it = c->txq.front();
+    size_t head_off = std::min(c->tx_front_written, it.head.size());
+    size_t ext_off = c->tx_front_written - head_off;
+    struct iovec iov[2];
+    int nio = 0;
+    if (head_off < it.head.size()) {
+      iov[nio].iov_base = it.head.data() + head_off;
+      iov[nio].iov_len = it.head.size() - head_off;
+      nio++;
+    }
+    if (it.ext && ext_off < it.ext_len) {
+      iov[nio].iov_base = (void*)(it.ext + ext_off);
+      iov[nio].iov_len = it.ext_len - ext_off;
+      nio++;
+    }
+    if (nio == 0) {
+      // fully written
+    } else {
+      ssize_t n = ::writev(c->fd, iov, nio);
+      if (n < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+        on_conn_dead(c);
+        return;
+      }
+      did_work = true;
+      c->tx_front_written += (size_t)n;
+    }
+    if (c->tx_front_written >= it.head.size() + it.ext_len) {
+      if (it.has_keepalive) {
+        dead_objs_.push_back(std::move(it.keepalive));
+        it.has_keepalive = false;
+      }
+      c->txq.pop_front();
+      c->tx_front_written = 0;
+    } else {
+      return;  // kernel buffer full
+    }
+  }
+}
+
+// ---- flush ----------------------------------------------------------------
+
+void Engine::check_flush_progress(Connection* c, uint64_t acked_flush_id) {
+  for (size_t i = 0; i < pending_flushes_.size();) {
+    Op* f = pending_flushes_[i];
+    f->flush_acks_pending.erase({c, acked_flush_id});
+    if (f->flush_acks_pending.empty() && f->flush_ops_pending.empty()) {
+      Completion comp;
+      comp.kind = Completion::Kind::FlushDone;
+      comp.op = f;
+      complete(std::move(comp));
+      pending_flushes_.erase(pending_flushes_.begin() + i);
+    } else {
+      i++;
+    }
+  }
+}
+
+// ---- connection death -----------------------------------------------------
+
+void Engine::on_conn_dead(Connection* c) {
+  if (c->dead) return;
+  c->dead = true;
+  if (c->fd >= 0) {
+    ::close(c->fd);
+    c->fd = -1;
+  }
+  // Reference behavior: the server keeps the (stale) endpoint entry in
+  // list_clients after the client closes (tests/test_basic.py:43-58).
+  if (c->ep) c->ep->conn = nullptr;
+  // A matched recv whose message was mid-stream goes back to pending: the
+  // data is lost but the recv must NOT complete (flush-semantics tests).
+  if (c->rx_recv_op) {
+    posted_recvs_.push_front(c->rx_recv_op);
+    c->rx_recv_op = nullptr;
+  }
+  if (c->rx_unexp) {
+    // Incomplete unexpected message: drop it.
+    for (auto it = unexpected_.begin(); it != unexpected_.end(); ++it) {
+      if (it->get() == c->rx_unexp) {
+        if ((*it)->bound_recv) posted_recvs_.push_front((*it)->bound_recv);
+        unexpected_.erase(it);
+        break;
+      }
+    }
+    c->rx_unexp = nullptr;
+  }
+  // Drop queued tx (deferring py keepalive refs).
+  for (auto& item : c->txq) {
+    if (item.has_keepalive) {
+      dead_objs_.push_back(std::move(item.keepalive));
+      item.has_keepalive = false;
+    }
+  }
+  c->txq.clear();
+  c->tx_front_written = 0;
+  // Flushes waiting on this conn can never complete: fail them.
+  for (size_t i = 0; i < pending_flushes_.size();) {
+    Op* f = pending_flushes_[i];
+    bool hit = false;
+    for (auto it = f->flush_acks_pending.begin();
+         it != f->flush_acks_pending.end();) {
+      if (it->first == c) {
+        it = f->flush_acks_pending.erase(it);
+        hit = true;
+      } else {
+        ++it;
+      }
+    }
+    if (hit) {
+      fail_op(f, "flush failed: connection reset");
+      pending_flushes_.erase(pending_flushes_.begin() + i);
+    } else {
+      i++;
+    }
+  }
+  // GPU sends routed to this conn will never be acked.
+  for (auto it = gpu_sends_.begin(); it != gpu_sends_.end();) {
+    if (it->second->conn == c) {
+      fail_op(it->second, "send failed: connection reset");
+      it = gpu_sends_.erase(it);
+    } else {
+      ++it;
+    }
+  }
+  if (mode_ == ClientMode && status_.load() == 2) {
+    // Our single connection died: subsequent ops fail, object stays
+    // closable.
+  }
+}
+
+// ---- completion plumbing --------------------------------------------------
+
+void Engine::complete(Completion&& comp) {
+  completions_.push_back(std::move(comp));
+}
+
+void Engine::fail_op(Op* op, const std::string& reason) {
+  Completion comp;
+  comp.kind = Completion::Kind::Fail;
+  comp.op = op;
+  comp.error = reason;
+  complete(std::move(comp));
+}
+
+void Engine::fire_completions() {
+  if (completions_.empty() && dead_objs_.empty()) return;
+  std::vector<Completion> comps;
+  comps.swap(completions_);
+  std::vector<py::object> dead;
+  dead.swap(dead_objs_);
+  py::gil_scoped_acquire gil;
+  dead.clear();
+  for (auto& comp : comps) {
+    try {
+      switch (comp.kind) {
+        case Completion::Kind::SendDone:
+        case Completion::Kind::FlushDone:
+          if (comp.op->done_cb.ptr()) comp.op->done_cb();
+          break;
+        case Completion::Kind::RecvDone:
+          if (comp.op->done_cb.ptr()) comp.op->done_cb(comp.a, comp.b);
+          break;
+        case Completion::Kind::Fail:
+          if (comp.op && comp.op->fail_cb.ptr()) comp.op->fail_cb(comp.error);
+          break;
+        case Completion::Kind::Accept:
+          if (accept_cb_.ptr()) accept_cb_(comp.ep);
+          break;
+        case Completion::Kind::Close:
+          if (comp.cb0.ptr()) comp.cb0();
+          break;
+        default:
+          break;
+      }
+    } catch (py::error_already_set& e) {
+      e.discard_as_unraisable("starway callback");
+    }
+    delete comp.op;  // py members freed under this GIL hold
+    comp.op = nullptr;
+    comp.cb0 = py::object();
+    comp.ep.reset();
+  }
+}
+
+// ---- teardown -------------------------------------------------------------
+
+void Engine::teardown() {
+  // 1. Fail everything still queued from Python threads.
+  {
+    std::vector<Op*> cmds;
+    drain_commands(cmds);
+    for (Op* op : cmds) fail_op(op, "operation canceled (endpoint closing)");
+  }
+  // 2. Let in-flight GPU pulls finish (bounded; they are plain copies), then
+  //    complete them and best-effort ack.
+  auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(10);
+  while (!gpu_pulls_.empty() &&
+         std::chrono::steady_clock::now() < deadline) {
+    bool did = false;
+    poll_gpu(did);
+    if (!did) sched_yield();
+  }
+  for (auto& p : gpu_pulls_) {
+    fail_op(p->recv_op, "operation canceled (endpoint closing)");
+    gpu::free_ticket(p->ticket);
+  }
+  gpu_pulls_.clear();
+  // 3. Cancel in-flight data: a connection with undelivered EAGER/RTS bytes
+  //    queued is closed abortively — close without flush loses in-flight
+  //    sends (the reference's delivery contract, tests/test_basic.py:250-278;
+  //    UCX analog: ucp_request_cancel on posted sends). This also returns any
+  //    mid-stream matched recv to the posted list so the next step cancels
+  //    it. Connections with only control frames left get a BYE and a short
+  //    drain below so peers see a clean shutdown.
+  for (auto& c : conns_) {
+    if (c->dead) continue;
+    bool has_data = c->tx_front_written > 0 && !c->txq.empty() &&
+                    c->txq.front().is_data;
+    for (auto& item : c->txq)
+      if (item.is_data) has_data = true;
+    if (has_data) {
+      on_conn_dead(c.get());
+    } else {
+      enqueue_frame(c.get(), FT_BYE, 0, 0, 0, nullptr, 0, false);
+    }
+  }
+  // 4. Cancel pending recvs / flushes / unacked GPU sends
+  //    (reference cancel contract: error string contains "cancel",
+  //    main.cpp:680-701, tests/test_basic.py:638-663).
+  for (Op* r : posted_recvs_) fail_op(r, "operation canceled (endpoint closing)");
+  posted_recvs_.clear();
+  for (auto& um : unexpected_)
+    if (um->bound_recv)
+      fail_op(um->bound_recv, "operation canceled (endpoint closing)");
+  unexpected_.clear();
+  for (Op* f : pending_flushes_) fail_op(f, "operation canceled (endpoint closing)");
+  pending_flushes_.clear();
+  for (auto& [id, op] : gpu_sends_)
+    fail_op(op, "operation canceled (endpoint closing)");
+  gpu_sends_.clear();
+  // 5. Best-effort drain of remaining control bytes (acks, BYE) ~200 ms.
+  deadline = std::chrono::steady_clock::now() + std::chrono::milliseconds(200);
+  while (std::chrono::steady_clock::now() < deadline) {
+    bool any = false;
+    for (auto& c : conns_)
+      if (!c->dead && c->want_write()) any = true;
+    if (!any) break;
+    bool did = false;
+    poll_sockets(5, did);
+  }
+  for (auto& c : conns_) {
+    if (c->fd >= 0) {
+      ::close(c->fd);
+      c->fd = -1;
+    }
+    if (c->ep) c->ep->conn = nullptr;
+    for (auto& item : c->txq)
+      if (item.has_keepalive) {
+        dead_objs_.push_back(std::move(item.keepalive));
+        item.has_keepalive = false;
+      }
+    c->txq.clear();
+  }
+  if (listen_fd_ >= 0) {
+    ::close(listen_fd_);
+    listen_fd_ = -1;
+  }
+  // 5. Fire all cancellations, then the close callback, then status 4
+  //    (ordering contract of reference main.cpp:469-549).
+  if (close_cb_.ptr()) {
+    Completion comp;
+    comp.kind = Completion::Kind::Close;
+    comp.cb0 = std::move(close_cb_);
+    complete(std::move(comp));
+  }
+  fire_completions();
+  {
+    py::gil_scoped_acquire gil;
+    accept_cb_ = py::object();
+    close_cb_ = py::object();
+    connect_cb_ = py::object();
+  }
+  status_.store(4, std::memory_order_release);
+}
+
+}  // namespace sw

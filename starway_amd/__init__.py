@@ -1,0 +1,309 @@
+"""starway_amd — MI355X-native async zero-copy tagged messaging for Python.
+
+A brand-new implementation with the capability surface of Clouder0/starway
+(reference: /root/reference/src/starway/__init__.py — API contract only).
+Where the reference wrapped OpenUCX, this library ships its own transport:
+
+* TCP control + CPU data plane with UCX-style eager semantics
+* GPU data plane: tagged send/recv of HIP device tensors, zero-copy over
+  xGMI via hipIpc handle exchange and hand-written gfx950 copy kernels
+* completion loop: per-object C++ progress thread polling hipEvents, with
+  asyncio bridging via ``loop.call_soon_threadsafe``
+
+Public surface (parity with the reference): ``Server``, ``Client``,
+``ServerEndpoint``, ``check_sys_libs``, ``list_benchmark_scenarios``.
+"""
+from __future__ import annotations
+
+import asyncio
+from collections.abc import Callable
+from typing import Any, Literal
+
+try:
+    from ._core import Client as _Client
+    from ._core import Context, ServerEndpoint
+    from ._core import Server as _Server
+    from ._core import gpu_available, gpu_device_count
+except ImportError as exc:  # pragma: no cover - build guidance
+    raise ImportError(
+        "starway_amd._core is not built. Run `python build_ext.py` (or "
+        "`python -c 'import __graft_entry__ as g; g.build()'`) from the repo "
+        "root first."
+    ) from exc
+
+from .benchmarks import list_scenarios as list_benchmark_scenarios  # noqa: E402
+
+
+def check_sys_libs() -> Literal["system"] | Literal["wheel"]:
+    """API-compat shim: the reference reported which UCX .so was loaded
+    (system vs wheel). Our transport is built in; report "system"."""
+    return "system"
+
+
+_context = Context()
+
+
+def _norm_buffer(buf: Any) -> Any:
+    """Normalize a message buffer for the native core.
+
+    Accepts numpy arrays / anything with the buffer protocol (host path) and
+    torch tensors: HIP tensors pass through via __cuda_array_interface__;
+    CPU torch tensors are viewed as numpy (zero-copy).
+    """
+    mod = type(buf).__module__
+    if mod.startswith("torch"):
+        if buf.is_cuda:
+            return buf
+        return buf.numpy()
+    return buf
+
+
+class Server:
+    def __init__(self) -> None:
+        self._server = _Server(_context)
+
+    def listen(self, addr: str, port: int) -> None:
+        self._server.listen(addr, port)
+
+    def listen_address(self) -> bytes:
+        self._server.listen_address()
+        return self.get_worker_address()
+
+    def set_accept_cb(self, on_accept: Callable[[ServerEndpoint], None]) -> None:
+        self._server.set_accept_callback(on_accept)
+
+    def get_worker_address(self) -> bytes:
+        return self._server.get_worker_address()
+
+    def list_clients(self):
+        return self._server.list_clients()
+
+    def aclose(self, loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def close_cb() -> None:
+            loop.call_soon_threadsafe(ret.set_result, None)
+
+        self._server.close(close_cb)
+        return ret
+
+    # -- raw callback API ---------------------------------------------------
+
+    def send(self, client_ep, buffer, tag, done_callback, fail_callback):
+        return self._server.send(
+            client_ep, _norm_buffer(buffer), tag, done_callback, fail_callback
+        )
+
+    def recv(self, buffer, tag, tag_mask, done_callback, fail_callback):
+        return self._server.recv(
+            _norm_buffer(buffer), tag, tag_mask, done_callback, fail_callback
+        )
+
+    def flush(self, done_callback, fail_callback):
+        return self._server.flush(done_callback, fail_callback)
+
+    def flush_ep(self, client_ep, done_callback, fail_callback):
+        return self._server.flush_ep(client_ep, done_callback, fail_callback)
+
+    # -- asyncio API --------------------------------------------------------
+
+    def asend(self, client_ep, buffer, tag: int,
+              loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def ok() -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._server.send(client_ep, _norm_buffer(buffer), tag, ok, bad)
+        return ret
+
+    def arecv(self, buffer, tag: int, tag_mask: int,
+              loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[tuple[int, int]] = asyncio.Future(loop=loop)
+
+        def ok(sender_tag: int, length: int) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret,
+                                                (sender_tag, length))
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._server.recv(_norm_buffer(buffer), tag, tag_mask, ok, bad)
+        return ret
+
+    def aflush(self, loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def ok() -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._server.flush(ok, bad)
+        return ret
+
+    def aflush_ep(self, client_ep,
+                  loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def ok() -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._server.flush_ep(client_ep, ok, bad)
+        return ret
+
+    def evaluate_perf(self, client_ep, msg_size: int) -> float:
+        return self._server.evaluate_perf(client_ep, msg_size)
+
+
+class Client:
+    def __init__(self) -> None:
+        self._client = _Client(_context)
+
+    def aconnect(self, addr: str, port: int,
+                 loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def connection_cb(status: str) -> None:
+            if status == "":
+                loop.call_soon_threadsafe(_set_result, ret, None)
+            else:
+                loop.call_soon_threadsafe(_set_exception, ret, status)
+
+        self._client.connect(addr, port, connection_cb)
+        return ret
+
+    def aconnect_address(self, remote_address: bytes,
+                         loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def connection_cb(status: str) -> None:
+            if status == "":
+                loop.call_soon_threadsafe(_set_result, ret, None)
+            else:
+                loop.call_soon_threadsafe(_set_exception, ret, status)
+
+        self._client.connect_address(remote_address, connection_cb)
+        return ret
+
+    def get_worker_address(self) -> bytes:
+        return self._client.get_worker_address()
+
+    def aclose(self, loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def close_cb() -> None:
+            loop.call_soon_threadsafe(_set_result, ret, None)
+
+        self._client.close(close_cb)
+        return ret
+
+    # -- raw callback API ---------------------------------------------------
+
+    def send(self, buffer, tag, done_callback, fail_callback):
+        return self._client.send(
+            _norm_buffer(buffer), tag, done_callback, fail_callback
+        )
+
+    def recv(self, buffer, tag, tag_mask, done_callback, fail_callback):
+        return self._client.recv(
+            _norm_buffer(buffer), tag, tag_mask, done_callback, fail_callback
+        )
+
+    def flush(self, done_callback, fail_callback):
+        return self._client.flush(done_callback, fail_callback)
+
+    # -- asyncio API --------------------------------------------------------
+
+    def asend(self, buffer, tag: int,
+              loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def ok() -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._client.send(_norm_buffer(buffer), tag, ok, bad)
+        return ret
+
+    def arecv(self, buffer, tag: int, tag_mask: int,
+              loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[tuple[int, int]] = asyncio.Future(loop=loop)
+
+        def ok(sender_tag: int, length: int) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret,
+                                                (sender_tag, length))
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._client.recv(_norm_buffer(buffer), tag, tag_mask, ok, bad)
+        return ret
+
+    def aflush(self, loop: asyncio.AbstractEventLoop | None = None):
+        if loop is None:
+            loop = asyncio.get_running_loop()
+        ret: asyncio.Future[None] = asyncio.Future(loop=loop)
+
+        def ok() -> None:
+            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+
+        def bad(reason: str) -> None:
+            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+
+        self._client.flush(ok, bad)
+        return ret
+
+    def evaluate_perf(self, msg_size: int) -> float:
+        return self._client.evaluate_perf(msg_size)
+
+
+def _set_result(fut: asyncio.Future, value) -> None:
+    if not fut.done():
+        fut.set_result(value)
+
+
+def _set_exception(fut: asyncio.Future, reason: str) -> None:
+    if not fut.done():
+        fut.set_exception(Exception(reason))
+
+
+__all__ = [
+    "Server",
+    "Client",
+    "ServerEndpoint",
+    "Context",
+    "check_sys_libs",
+    "list_benchmark_scenarios",
+    "gpu_available",
+    "gpu_device_count",
+]
