@@ -1,0 +1,253 @@
+#!/usr/bin/env python3
+"""Driver-contract benchmark for starway_amd.
+
+Headline metric (BASELINE.json): aggregate GB/s of tagged all-pairs
+send/recv of HIP device tensors over xGMI, at N MI355X endpoints
+(one process per GPU), plus 64 B pingpong half-RTT latency.
+
+One "step" = every rank tag-sends a fixed per-rank byte volume
+(``--msg-bytes``, default 256 MiB), split evenly across all peers, and
+receives the matching inbound messages (weak scaling: per-GPU work is fixed
+as N grows). N=1 runs a same-process loopback pair on cuda:0 (HBM-bound);
+N>=2 is xGMI-bound. With no CUDA device the bench falls back to CPU numpy
+buffers over localhost TCP (small default size) so it stays runnable
+anywhere.
+
+Launch (driver): N=1 plain; N>1 via
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+torch.distributed (gloo) is used ONLY for rank rendezvous (worker-address
+exchange + barriers); the data plane is starway_amd itself.
+
+Rank 0 prints ONE JSON line with the aggregate result.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import numpy as np  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description="starway_amd driver benchmark")
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--msg-bytes", type=int, default=256 * 1024 * 1024,
+                   help="per-rank outbound bytes per step (split across peers)")
+    p.add_argument("--lat-iters", type=int, default=200,
+                   help="64B pingpong iterations for the latency probe")
+    p.add_argument("--device", default=None,
+                   help="force 'cpu' or 'cuda' (default: auto)")
+    p.add_argument("--json-out", default=None)
+    return p.parse_args()
+
+
+def make_tag(src_rank: int, step: int) -> int:
+    return (1 << 60) | (src_rank << 32) | (step & 0xFFFFFFFF)
+
+
+LAT_TAG_BASE = 1 << 59
+
+
+async def run_rank(args, rank: int, world: int, device: str, dist):
+    import starway_amd as sw
+
+    full_mask = (1 << 64) - 1
+    n_peers = max(1, world - 1)
+    per_peer = args.msg_bytes // n_peers if world > 1 else args.msg_bytes
+
+    def alloc(n, fill=None):
+        if device == "cpu":
+            a = np.empty(n, dtype=np.uint8)
+            if fill is not None:
+                a.fill(fill)
+            return a
+        import torch
+
+        t = torch.empty(n, dtype=torch.uint8, device="cuda")
+        if fill is not None:
+            t.fill_(fill)
+        return t
+
+    def sync_device():
+        if device != "cpu":
+            import torch
+
+            torch.cuda.synchronize()
+
+    server = sw.Server()
+    blob = server.listen_address()
+
+    if world > 1:
+        blobs = [None] * world
+        dist.all_gather_object(blobs, blob)
+        # rank i's client[j] connects to rank j's server (full mesh).
+        clients = {}
+        for j in range(world):
+            if j == rank:
+                continue
+            c = sw.Client()
+            await c.aconnect_address(blobs[j])
+            clients[j] = c
+        dist.barrier()
+        peers = sorted(clients.keys())
+    else:
+        c = sw.Client()
+        await c.aconnect_address(blob)
+        clients = {0: c}
+        peers = [0]
+
+    send_bufs = {j: alloc(per_peer, fill=(rank * 31 + j) % 251) for j in peers}
+    recv_bufs = {j: alloc(per_peer) for j in peers}
+    sync_device()
+
+    async def step(step_idx: int):
+        recvs = [
+            server.arecv(recv_bufs[j], make_tag(j, step_idx), full_mask)
+            for j in peers
+        ]
+        sends = [
+            clients[j].asend(send_bufs[j], make_tag(rank, step_idx))
+            for j in peers
+        ]
+        await asyncio.gather(*sends, *recvs)
+
+    # ---- warmup ----
+    for i in range(args.warmup):
+        await step(i)
+    sync_device()
+    if world > 1:
+        dist.barrier()
+
+    # ---- timed region ----
+    t0 = time.perf_counter()
+    for i in range(args.warmup, args.warmup + args.steps):
+        await step(i)
+    sync_device()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        import torch
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    # ---- 64 B pingpong latency probe (rank0 <-> rank1, or loopback) ----
+    half_rtt_us = None
+    lat_samples = []
+    ping = alloc(64, fill=1)
+    pong = alloc(64)
+    if world == 1:
+        for i in range(args.lat_iters):
+            t1 = time.perf_counter()
+            fut = server.arecv(pong, LAT_TAG_BASE + i, full_mask)
+            await clients[0].asend(ping, LAT_TAG_BASE + i)
+            await fut
+            lat_samples.append(time.perf_counter() - t1)
+    elif rank in (0, 1):
+        other = 1 - rank
+        for i in range(args.lat_iters):
+            if rank == 0:
+                t1 = time.perf_counter()
+                fut = server.arecv(pong, make_tag(other, 1 << 30 | i), full_mask)
+                await clients[other].asend(ping, make_tag(rank, 1 << 30 | i))
+                await fut
+                lat_samples.append(time.perf_counter() - t1)
+            else:
+                await server.arecv(pong, make_tag(other, 1 << 30 | i), full_mask)
+                await clients[other].asend(ping, make_tag(rank, 1 << 30 | i))
+    if lat_samples:
+        half_rtt_us = float(np.percentile(np.array(lat_samples) * 1e6, 50)) / 2.0
+
+    if world > 1:
+        dist.barrier()
+
+    # ---- teardown ----
+    for c in clients.values():
+        await c.aclose()
+    await server.aclose()
+
+    return elapsed, half_rtt_us
+
+
+def main() -> int:
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    device = args.device
+    if device is None:
+        try:
+            import torch
+
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        except ImportError:
+            device = "cpu"
+    if device == "cpu" and args.msg_bytes > 64 * 1024 * 1024:
+        # CPU/TCP fallback: keep the default run snappy.
+        args.msg_bytes = 16 * 1024 * 1024
+
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        if device == "cuda":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group("gloo")
+
+    elapsed, half_rtt_us = asyncio.run(run_rank(args, rank, world, device, dist))
+
+    if rank == 0:
+        total_bytes = world * args.msg_bytes * args.steps
+        gbps = total_bytes / elapsed / 1e9
+        result = {
+            "metric": "tagged_allpairs_bandwidth",
+            "value": round(gbps, 3),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "uint8",
+            "data": "synthetic",
+            "config": {
+                "model": "tagged-pingpong-allpairs",
+                "message_bytes_per_rank": args.msg_bytes,
+                "message_bytes_per_peer": args.msg_bytes // max(1, world - 1),
+                "endpoints": world,
+                "device": device,
+                "parallelism": f"p2p-mesh{world}",
+                "pingpong_64B_half_rtt_us":
+                    round(half_rtt_us, 2) if half_rtt_us else None,
+                "note": "N=1 is same-GPU loopback (HBM-bound); N>=2 is "
+                        "xGMI-bound; value counts each sent byte once",
+            },
+        }
+        line = json.dumps(result)
+        print(line, flush=True)
+        if args.json_out:
+            Path(args.json_out).write_text(line + "\n")
+
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

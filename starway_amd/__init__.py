@@ -43,17 +43,45 @@ def check_sys_libs() -> Literal["system"] | Literal["wheel"]:
 _context = Context()
 
 
+class _CudaShim:
+    """Zero-copy view of a torch HIP tensor for the native core.
+
+    Exposes __cuda_array_interface__ built from data_ptr()/nbytes, pinning
+    the tensor via a strong reference (the core keeps this shim alive as the
+    op's keepalive). Used instead of the tensor's own CAI so behavior is
+    identical across torch builds.
+    """
+
+    __slots__ = ("_t", "__cuda_array_interface__")
+
+    def __init__(self, t) -> None:
+        if not t.is_contiguous():
+            raise ValueError(
+                "device message buffers must be contiguous for zero-copy "
+                "transfer (call .contiguous() first)"
+            )
+        self._t = t
+        nbytes = t.numel() * t.element_size()
+        self.__cuda_array_interface__ = {
+            "data": (t.data_ptr(), False),
+            "shape": (nbytes,),
+            "typestr": "|u1",
+            "strides": None,
+            "version": 2,
+        }
+
+
 def _norm_buffer(buf: Any) -> Any:
     """Normalize a message buffer for the native core.
 
     Accepts numpy arrays / anything with the buffer protocol (host path) and
-    torch tensors: HIP tensors pass through via __cuda_array_interface__;
-    CPU torch tensors are viewed as numpy (zero-copy).
+    torch tensors: HIP tensors go zero-copy via a __cuda_array_interface__
+    shim; CPU torch tensors are viewed as numpy (zero-copy).
     """
     mod = type(buf).__module__
     if mod.startswith("torch"):
         if buf.is_cuda:
-            return buf
+            return _CudaShim(buf)
         return buf.numpy()
     return buf
 

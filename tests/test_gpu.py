@@ -1,0 +1,336 @@
+"""GPU-path tests (MI355X): gfx950 copy-kernel numerics, device-tensor
+tagged messaging (same-process raw-ptr path and cross-process hipIpc path),
+mixed host/device transfers, flush-with-GPU-sends, truncation RECV_FAIL.
+
+Numerics policy: message delivery is a byte-exact copy, so every test
+compares against the untouched source tensor (the PyTorch reference of a
+copy op is the tensor itself).
+"""
+import asyncio
+import contextlib
+import multiprocessing as mp
+import os
+import random
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+pytest.importorskip("torch")
+import torch  # noqa: E402
+
+import starway_amd as sw  # noqa: E402
+from starway_amd import _core  # noqa: E402
+
+
+@pytest.fixture
+def port():
+    return random.randint(10000, 50000)
+
+
+@contextlib.asynccontextmanager
+async def loopback():
+    server = sw.Server()
+    client = sw.Client()
+    addr = server.listen_address()
+    await client.aconnect_address(addr)
+    try:
+        yield server, client
+    finally:
+        await client.aclose()
+        await server.aclose()
+
+
+# =============================================================================
+# Copy-kernel numerics (direct)
+# =============================================================================
+
+
+@pytest.mark.parametrize("nbytes", [1, 16, 4096, 1 << 20, (1 << 20) + 13])
+def test_copy_kernel_exact(nbytes):
+    src = torch.randint(0, 256, (nbytes,), dtype=torch.uint8, device="cuda")
+    dst = torch.zeros_like(src)
+    torch.cuda.synchronize()
+    _core._copy_device_sync(dst.data_ptr(), src.data_ptr(), nbytes, 0)
+    assert torch.equal(src, dst)
+
+
+def test_copy_kernel_unaligned_slices():
+    base = torch.randint(0, 256, (1 << 16,), dtype=torch.uint8, device="cuda")
+    out = torch.zeros_like(base)
+    torch.cuda.synchronize()
+    # co-aligned but offset from the 16B boundary
+    for off, ln in [(1, 4097), (3, 1 << 14), (7, 255), (15, 16)]:
+        _core._copy_device_sync(
+            out.data_ptr() + off, base.data_ptr() + off, ln, 0
+        )
+        assert torch.equal(base[off : off + ln], out[off : off + ln])
+
+
+def test_copy_kernel_large_nt_path():
+    # >64 MiB goes through the non-temporal kernel variant.
+    n = 96 << 20
+    src = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+    dst = torch.zeros_like(src)
+    torch.cuda.synchronize()
+    _core._copy_device_sync(dst.data_ptr(), src.data_ptr(), n, 0)
+    assert torch.equal(src, dst)
+
+
+# =============================================================================
+# Same-process device messaging (raw-ptr RTS path)
+# =============================================================================
+
+
+@pytest.mark.parametrize("nbytes", [64, 4096, 1 << 20, 64 << 20])
+async def test_device_send_recv_same_process(nbytes):
+    async with loopback() as (server, client):
+        src = torch.randint(0, 256, (nbytes,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        fut = server.arecv(dst, 0, 0)
+        await client.asend(src, 5)
+        tag, length = await fut
+        torch.cuda.synchronize()
+        assert tag == 5 and length == nbytes
+        assert torch.equal(src, dst)
+
+
+async def test_device_send_before_recv_posted():
+    # RTS lands in the unexpected queue; matched when the recv is posted.
+    async with loopback() as (server, client):
+        src = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        send_fut = client.asend(src, 77)
+        await asyncio.sleep(0.05)
+        tag, length = await server.arecv(dst, 77, (1 << 64) - 1)
+        await send_fut
+        torch.cuda.synchronize()
+        assert tag == 77 and torch.equal(src, dst)
+
+
+async def test_device_server_to_client():
+    async with loopback() as (server, client):
+        ep = next(iter(server.list_clients()))
+        src = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        fut = client.arecv(dst, 0, 0)
+        await asyncio.sleep(0.01)
+        await server.asend(ep, src, 9)
+        tag, _ = await fut
+        torch.cuda.synchronize()
+        assert tag == 9 and torch.equal(src, dst)
+
+
+async def test_device_duplex_concurrent():
+    async with loopback() as (server, client):
+        ep = next(iter(server.list_clients()))
+        n = 8 << 20
+        a = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        b = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        ra = torch.zeros_like(a)
+        rb = torch.zeros_like(b)
+        torch.cuda.synchronize()
+        futs = await asyncio.gather(
+            server.arecv(ra, 1, (1 << 64) - 1),
+            client.arecv(rb, 2, (1 << 64) - 1),
+            client.asend(a, 1),
+            server.asend(ep, b, 2),
+        )
+        torch.cuda.synchronize()
+        assert futs[0][0] == 1 and futs[1][0] == 2
+        assert torch.equal(a, ra) and torch.equal(b, rb)
+
+
+# =============================================================================
+# Mixed host/device
+# =============================================================================
+
+
+async def test_gpu_to_cpu_recv():
+    async with loopback() as (server, client):
+        src = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device="cuda")
+        dst = np.zeros(1 << 20, dtype=np.uint8)
+        torch.cuda.synchronize()
+        fut = server.arecv(dst, 0, 0)
+        await client.asend(src, 3)
+        tag, length = await fut
+        assert tag == 3 and length == 1 << 20
+        np.testing.assert_array_equal(src.cpu().numpy(), dst)
+
+
+async def test_cpu_to_gpu_recv():
+    async with loopback() as (server, client):
+        src = np.random.randint(0, 256, 1 << 20, dtype=np.uint8)
+        dst = torch.zeros(1 << 20, dtype=torch.uint8, device="cuda")
+        fut = server.arecv(dst, 0, 0)
+        await asyncio.sleep(0.01)
+        await client.asend(src, 4)
+        tag, length = await fut
+        torch.cuda.synchronize()
+        assert tag == 4 and length == 1 << 20
+        np.testing.assert_array_equal(src, dst.cpu().numpy())
+
+
+# =============================================================================
+# Flush / cancel / truncation semantics with GPU sends
+# =============================================================================
+
+
+async def test_gpu_send_completion_means_delivery():
+    async with loopback() as (server, client):
+        src = torch.full((1 << 16,), 7, dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        fut = server.arecv(dst, 0, 0)
+        await client.asend(src, 1)  # completes only after RECV_DONE
+        await fut
+        torch.cuda.synchronize()
+        assert torch.equal(src, dst)
+        await client.aflush()  # nothing pending; must resolve
+
+
+async def test_gpu_truncation_fails_both_sides():
+    async with loopback() as (server, client):
+        src = torch.zeros(4096, dtype=torch.uint8, device="cuda")
+        dst = torch.zeros(128, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        recv_fut = server.arecv(dst, 0, 0)
+        send_fut = client.asend(src, 1)
+        with pytest.raises(Exception, match="truncated"):
+            await recv_fut
+        with pytest.raises(Exception, match="truncated"):
+            await send_fut
+
+
+async def test_gpu_pending_send_canceled_on_close():
+    server = sw.Server()
+    client = sw.Client()
+    addr = server.listen_address()
+    await client.aconnect_address(addr)
+    src = torch.zeros(1 << 20, dtype=torch.uint8, device="cuda")
+    torch.cuda.synchronize()
+    send_fut = client.asend(src, 123)  # no recv posted: RTS never acked
+    await asyncio.sleep(0.05)
+    close_fut = client.aclose()
+    with pytest.raises(Exception, match="cancel|reset"):
+        await send_fut
+    await close_fut
+    await server.aclose()
+
+
+# =============================================================================
+# Cross-process hipIpc path (two processes, one GPU)
+# =============================================================================
+
+
+def _ipc_child_server(port: int, nbytes: int):
+    import torch
+
+    import starway_amd as sw
+
+    async def inner():
+        server = sw.Server()
+        server.listen("127.0.0.1", port)
+        connected = asyncio.Event()
+        loop = asyncio.get_running_loop()
+        server.set_accept_cb(lambda ep: loop.call_soon_threadsafe(connected.set))
+        await connected.wait()
+        ep = next(iter(server.list_clients()))
+        src = torch.arange(nbytes, dtype=torch.uint8, device="cuda") % 251
+        src = src.contiguous()
+        torch.cuda.synchronize()
+        await server.asend(ep, src, 21)  # completes on RECV_DONE (delivery)
+        await server.aflush_ep(ep)
+        await server.aclose()
+
+    asyncio.run(inner())
+
+
+async def test_cross_process_ipc_device_transfer(port):
+    nbytes = 32 << 20
+    ctx = mp.get_context("spawn")
+    p = ctx.Process(target=_ipc_child_server, args=(port, nbytes))
+    p.start()
+    try:
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+        dst = torch.zeros(nbytes, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        tag, length = await client.arecv(dst, 0, 0)
+        torch.cuda.synchronize()
+        assert tag == 21 and length == nbytes
+        expect = torch.arange(nbytes, dtype=torch.uint8, device="cuda") % 251
+        assert torch.equal(dst, expect)
+        await client.aclose()
+    finally:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
+
+
+def _ipc_child_client(port: int, nbytes: int):
+    import torch
+
+    import starway_amd as sw
+
+    async def inner():
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+        src = (torch.arange(nbytes, dtype=torch.uint8, device="cuda") * 3) % 241
+        src = src.contiguous()
+        torch.cuda.synchronize()
+        await client.asend(src, 33)
+        await client.aflush()
+        await client.aclose()
+
+    asyncio.run(inner())
+
+
+async def test_cross_process_ipc_client_to_server(port):
+    nbytes = 8 << 20
+    server = sw.Server()
+    server.listen("127.0.0.1", port)
+    ctx = mp.get_context("spawn")
+    p = ctx.Process(target=_ipc_child_client, args=(port, nbytes))
+    p.start()
+    try:
+        dst = torch.zeros(nbytes, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        tag, length = await server.arecv(dst, 0, 0)
+        torch.cuda.synchronize()
+        assert tag == 33 and length == nbytes
+        expect = (torch.arange(nbytes, dtype=torch.uint8, device="cuda") * 3) % 241
+        assert torch.equal(dst, expect)
+    finally:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
+        await server.aclose()
+
+
+# =============================================================================
+# Benchmark-path sanity: bench.py loopback step on GPU
+# =============================================================================
+
+
+async def test_bench_loopback_step():
+    async with loopback() as (server, client):
+        n = 16 << 20
+        src = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        for step in range(3):
+            tag = (1 << 60) | step
+            fut = server.arecv(dst, tag, (1 << 64) - 1)
+            await client.asend(src, tag)
+            await fut
+        torch.cuda.synchronize()
+        assert torch.equal(src, dst)
