@@ -154,8 +154,10 @@ struct Op {
   uint64_t recv_len = 0;          // actual message length for recv completion
   uint64_t recv_sender_tag = 0;
   bool gpu_send_awaiting_ack = false;
-  // Flush bookkeeping:
-  std::set<std::pair<Connection*, uint64_t>> flush_acks_pending;
+  // Flush bookkeeping (UCX worker/ep-flush semantics: complete when all
+  // bytes queued before the flush are written to the wire AND all GPU
+  // rendezvous sends posted before it are delivered):
+  std::map<Connection*, uint64_t> flush_write_targets;
   std::set<uint64_t> flush_ops_pending;  // GPU send op ids in flight
 };
 
@@ -228,7 +230,8 @@ struct Connection {
   // --- tx ---
   std::deque<TxItem> txq;
   size_t tx_front_written = 0;
-  size_t tx_bytes_queued = 0;
+  uint64_t tx_enqueued_bytes = 0;  // lifetime totals (flush targets)
+  uint64_t tx_written_bytes = 0;
 
   bool want_write() const { return !txq.empty(); }
 };
@@ -318,7 +321,7 @@ class Engine {
   void send_hello(Connection* c);
   void on_hello(Connection* c);
   void on_conn_dead(Connection* c);
-  void check_flush_progress(Connection* c, uint64_t acked_flush_id);
+  void check_flush_progress();
   void on_gpu_send_acked(uint64_t op_id, bool failed, const std::string& err);
   void teardown();
   void fire_completions();  // acquires GIL, drains completions_
@@ -366,6 +369,7 @@ class Engine {
   bool connect_done_ = false;
 
   py::object accept_cb_;
+  std::mutex close_mu_;  // orders close_cb_ assignment vs teardown's read
   py::object close_cb_;
   bool have_accept_cb_ = false;
 

@@ -358,6 +358,10 @@ void Engine::set_accept_callback(py::object cb) {
 }
 
 void Engine::close(py::object cb) {
+  // The mutex makes the callback assignment visible to teardown(): the
+  // engine thread may observe status 3 immediately after the CAS, and must
+  // not read close_cb_ until the assignment below is complete.
+  std::lock_guard<std::mutex> lk(close_mu_);
   int expect = 2;
   if (!status_.compare_exchange_strong(expect, 3))
     throw std::runtime_error(
@@ -481,7 +485,11 @@ double Engine::evaluate_perf(std::shared_ptr<EndpointInfo> ep,
 void Engine::thread_main() {
   if (mode_ == ClientMode && connect_requested_) {
     do_connect_start();
-    if (status_.load() != 2) return;  // connect failed; callback already fired
+    // 4 => connect failed (callback already fired): exit without teardown.
+    // 3 can appear here when close() lands right after the successful
+    // connect published status 2 — fall through so teardown() runs and the
+    // close callback fires.
+    if (status_.load(std::memory_order_acquire) == 4) return;
   } else {
     status_.store(2, std::memory_order_release);
   }
@@ -635,6 +643,7 @@ void Engine::loop_iteration(bool& did_work) {
   if (!busy && idle_iters_ > 50000) timeout = 1;
   poll_sockets(timeout, did_work);
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
+  if (!pending_flushes_.empty()) check_flush_progress();
   if (!completions_.empty()) {
     fire_completions();
     did_work = true;
@@ -701,16 +710,15 @@ void Engine::process_command(Op* op) {
           if (!c->dead && c->hello_received) targets.push_back(c.get());
       }
       for (Connection* c : targets) {
-        uint64_t fid = next_flush_id_++;
-        enqueue_frame(c, FT_FLUSH_REQ, 0, fid, 0, nullptr, 0, false);
-        op->flush_acks_pending.insert({c, fid});
+        if (c->tx_written_bytes < c->tx_enqueued_bytes)
+          op->flush_write_targets[c] = c->tx_enqueued_bytes;
       }
       for (auto& [id, sop] : gpu_sends_) {
         if (op->type == OpType::Flush ||
             (!targets.empty() && sop->conn == targets[0]))
           op->flush_ops_pending.insert(id);
       }
-      if (op->flush_acks_pending.empty() && op->flush_ops_pending.empty()) {
+      if (op->flush_write_targets.empty() && op->flush_ops_pending.empty()) {
         Completion comp;
         comp.kind = Completion::Kind::FlushDone;
         comp.op = op;
@@ -906,8 +914,7 @@ void Engine::on_frame(Connection* c) {
       enqueue_frame(c, FT_FLUSH_ACK, 0, h.op_id, 0, nullptr, 0, true);
       return;
     case FT_FLUSH_ACK:
-      check_flush_progress(c, h.op_id);
-      return;
+      return;  // legacy; flushes complete on write totals
     case FT_BYE:
       on_conn_dead(c);
       return;
@@ -1187,7 +1194,7 @@ void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
   for (size_t i = 0; i < pending_flushes_.size();) {
     Op* f = pending_flushes_[i];
     f->flush_ops_pending.erase(op_id);
-    if (f->flush_acks_pending.empty() && f->flush_ops_pending.empty()) {
+    if (f->flush_write_targets.empty() && f->flush_ops_pending.empty()) {
       Completion comp;
       comp.kind = Completion::Kind::FlushDone;
       comp.op = f;
@@ -1282,6 +1289,7 @@ void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
   memcpy(item.head.data(), &h, sizeof(h));
   if (payload_len) memcpy(item.head.data() + sizeof(h), payload, payload_len);
   item.is_data = (t == FT_RTS);
+  c->tx_enqueued_bytes += item.head.size();
   if (priority && !c->txq.empty()) {
     // Insert at the first frame boundary, after any queued priority frames.
     size_t pos = c->tx_front_written > 0 ? 1 : 0;
@@ -1316,6 +1324,7 @@ void Engine::enqueue_eager(Connection* c, Op* op) {
     item.has_keepalive = true;
     item.keepalive = std::move(op->keepalive);  // moved, no refcount touch
   }
+  c->tx_enqueued_bytes += item.head.size() + item.ext_len;
   c->txq.push_back(std::move(item));
   bool dummy = false;
   handle_writable(c, dummy);
@@ -1349,6 +1358,7 @@ void Engine::handle_writable(Connection* c, bool& did_work) {
       }
       did_work = true;
       c->tx_front_written += (size_t)n;
+      c->tx_written_bytes += (uint64_t)n;
     }
     if (c->tx_front_written >= it.head.size() + it.ext_len) {
       if (it.has_keepalive) {
@@ -1365,11 +1375,17 @@ void Engine::handle_writable(Connection* c, bool& did_work) {
 
 // ---- flush ----------------------------------------------------------------
 
-void Engine::check_flush_progress(Connection* c, uint64_t acked_flush_id) {
+void Engine::check_flush_progress() {
   for (size_t i = 0; i < pending_flushes_.size();) {
     Op* f = pending_flushes_[i];
-    f->flush_acks_pending.erase({c, acked_flush_id});
-    if (f->flush_acks_pending.empty() && f->flush_ops_pending.empty()) {
+    for (auto it = f->flush_write_targets.begin();
+         it != f->flush_write_targets.end();) {
+      if (it->first->tx_written_bytes >= it->second)
+        it = f->flush_write_targets.erase(it);
+      else
+        ++it;
+    }
+    if (f->flush_write_targets.empty() && f->flush_ops_pending.empty()) {
       Completion comp;
       comp.kind = Completion::Kind::FlushDone;
       comp.op = f;
@@ -1419,17 +1435,17 @@ void Engine::on_conn_dead(Connection* c) {
   }
   c->txq.clear();
   c->tx_front_written = 0;
-  // Flushes waiting on this conn can never complete: fail them.
+  // Flushes whose write target on this conn was not reached can never
+  // complete: fail them.
   for (size_t i = 0; i < pending_flushes_.size();) {
     Op* f = pending_flushes_[i];
+    auto it = f->flush_write_targets.find(c);
     bool hit = false;
-    for (auto it = f->flush_acks_pending.begin();
-         it != f->flush_acks_pending.end();) {
-      if (it->first == c) {
-        it = f->flush_acks_pending.erase(it);
-        hit = true;
+    if (it != f->flush_write_targets.end()) {
+      if (c->tx_written_bytes >= it->second) {
+        f->flush_write_targets.erase(it);  // satisfied before death
       } else {
-        ++it;
+        hit = true;
       }
     }
     if (hit) {
@@ -1591,13 +1607,16 @@ void Engine::teardown() {
     ::close(listen_fd_);
     listen_fd_ = -1;
   }
-  // 5. Fire all cancellations, then the close callback, then status 4
+  // 6. Fire all cancellations, then the close callback, then status 4
   //    (ordering contract of reference main.cpp:469-549).
-  if (close_cb_.ptr()) {
-    Completion comp;
-    comp.kind = Completion::Kind::Close;
-    comp.cb0 = std::move(close_cb_);
-    complete(std::move(comp));
+  {
+    std::lock_guard<std::mutex> lk(close_mu_);
+    if (close_cb_.ptr()) {
+      Completion comp;
+      comp.kind = Completion::Kind::Close;
+      comp.cb0 = std::move(close_cb_);
+      complete(std::move(comp));
+    }
   }
   fire_completions();
   {

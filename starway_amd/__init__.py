@@ -16,8 +16,37 @@ Public surface (parity with the reference): ``Server``, ``Client``,
 from __future__ import annotations
 
 import asyncio
+import ctypes
+import os
 from collections.abc import Callable
+from importlib.util import find_spec
+from pathlib import Path
 from typing import Any, Literal
+
+# ---------------------------------------------------------------------------
+# HIP runtime loader policy (the analog of the reference's UCX loader,
+# reference src/starway/__init__.py:14-51): our native core NEEDs
+# libamdhip64.so.7. PyTorch-ROCm bundles its own copy of the runtime with
+# the same soname; if _core loads the system ROCm runtime first, a later
+# torch import binds to it too and mixed-version HSA state breaks device
+# discovery ("No HIP GPUs are available"). Default policy: when torch is
+# installed, preload torch's bundled runtime (RTLD_GLOBAL) so everyone
+# shares one copy. STARWAY_USE_SYSTEM_HIP=true opts into the system ROCm
+# runtime instead (only safe in torch-free processes).
+# ---------------------------------------------------------------------------
+_used_hip = "system"
+if os.environ.get("STARWAY_USE_SYSTEM_HIP", "false") != "true":
+    _torch_spec = find_spec("torch")
+    if _torch_spec and _torch_spec.origin:
+        _libdir = Path(_torch_spec.origin).parent / "lib"
+        for _name in ("libhsa-runtime64.so", "libamdhip64.so"):
+            _p = _libdir / _name
+            if _p.exists():
+                try:
+                    ctypes.CDLL(str(_p), mode=ctypes.RTLD_GLOBAL)
+                    _used_hip = "wheel"
+                except OSError:
+                    pass
 
 try:
     from ._core import Client as _Client
@@ -35,9 +64,11 @@ from .benchmarks import list_scenarios as list_benchmark_scenarios  # noqa: E402
 
 
 def check_sys_libs() -> Literal["system"] | Literal["wheel"]:
-    """API-compat shim: the reference reported which UCX .so was loaded
-    (system vs wheel). Our transport is built in; report "system"."""
-    return "system"
+    """Which HIP runtime the native core is bound to: "wheel" = the copy
+    bundled inside the installed torch package (default when torch is
+    present), "system" = /opt/rocm. The reference's analog reported which
+    UCX .so was loaded (reference src/starway/__init__.py:63-65)."""
+    return _used_hip  # type: ignore[return-value]
 
 
 _context = Context()
