@@ -200,6 +200,9 @@ struct UnexpectedMsg {
   size_t got = 0;
   bool complete = false;
   Op* bound_recv = nullptr;   // recv matched while message still streaming in
+  // When a host recv binds mid-stream, the staged prefix is copied into its
+  // buffer and the remaining bytes stream straight there (no double copy).
+  uint8_t* redirect_dst = nullptr;
 };
 
 struct Connection {

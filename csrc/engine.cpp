@@ -845,7 +845,9 @@ void Engine::handle_readable(Connection* c, bool& did_work) {
             dst = c->rx_recv_op->buf.ptr + done;
           }
         } else if (c->rx_unexp) {
-          dst = c->rx_unexp->data.data() + done;
+          dst = (c->rx_unexp->redirect_dst ? c->rx_unexp->redirect_dst
+                                           : c->rx_unexp->data.data()) +
+                done;
         } else {
           if (c->rx_discard.size() < (64 << 10)) c->rx_discard.resize(64 << 10);
           dst = c->rx_discard.data();
@@ -1227,6 +1229,12 @@ bool Engine::try_match_unexpected(Op* op) {
     }
     if (!um->complete) {
       um->bound_recv = op;  // delivered when the stream finishes
+      if (op->buf.device < 0 && um->size <= op->buf.size) {
+        // Redirect: staged prefix now, remainder streams zero-copy.
+        memcpy(op->buf.ptr, um->data.data(), um->got);
+        um->redirect_dst = op->buf.ptr;
+        um->data.clear();
+      }
       return true;
     }
     UnexpectedMsg* owned = it->release();
@@ -1239,6 +1247,16 @@ bool Engine::try_match_unexpected(Op* op) {
 
 void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
   std::unique_ptr<UnexpectedMsg> guard(um);
+  if (um->redirect_dst) {
+    // Stream was redirected into the recv buffer; payload already in place.
+    Completion comp;
+    comp.kind = Completion::Kind::RecvDone;
+    comp.op = op;
+    comp.a = um->tag;
+    comp.b = um->size;
+    complete(std::move(comp));
+    return;
+  }
   if (um->size > op->buf.size) {
     fail_op(op, "receive failed: message truncated (len " +
                     std::to_string(um->size) + " > buffer " +
