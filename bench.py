@@ -205,7 +205,9 @@ def main() -> int:
 
         dist = dist_mod
         if device == "cuda":
-            torch.cuda.set_device(local_rank)
+            # modulo so a 2-rank run on a 1-GPU box exercises the full
+            # cross-process hipIpc path (the driver's 8-GPU run has 1:1).
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group("gloo")
 
     elapsed, half_rtt_us = asyncio.run(run_rank(args, rank, world, device, dist))

@@ -1,0 +1,199 @@
+#!/usr/bin/env python3
+"""Message-size sweep for the BASELINE table: tagged pingpong GB/s and p50
+half-RTT from 64 B to 256 MB.
+
+Modes:
+  * single process (default): Server+Client loopback in one process; on a
+    GPU host buffers are HIP device tensors (same-GPU HBM copy path), else
+    CPU numpy over localhost TCP.
+  * --cross-process: the server lives in a spawned subprocess => the full
+    worker-address handshake + hipIpc import path, still on one GPU when
+    only one is visible.
+
+Writes a JSON + text table to --out (default gpurun_out/sweep.json).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import multiprocessing as mp
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import numpy as np  # noqa: E402
+
+SIZES = [64, 1024, 64 * 1024, 1 << 20, 16 << 20, 256 << 20]
+FULL = (1 << 64) - 1
+
+
+def pick_iters(size: int) -> tuple[int, int]:
+    if size <= 64 * 1024:
+        return 200, 1000
+    if size <= (16 << 20):
+        return 20, 100
+    return 5, 20
+
+
+def _alloc(n, device, fill=None):
+    if device == "cpu":
+        a = np.empty(n, dtype=np.uint8)
+        if fill is not None:
+            a.fill(fill)
+        return a
+    import torch
+
+    t = torch.empty(n, dtype=torch.uint8, device="cuda")
+    if fill is not None:
+        t.fill_(fill)
+    torch.cuda.synchronize()
+    return t
+
+
+def _server_proc(port: int, device: str, sizes: list[int]):
+    import starway_amd as sw
+
+    async def inner():
+        server = sw.Server()
+        server.listen("127.0.0.1", port)
+        for size in sizes:
+            warmup, iters = pick_iters(size)
+            rbuf = _alloc(size, device)
+            sbuf = _alloc(size, device, fill=1)
+            ep = None
+            for i in range(warmup + iters):
+                await server.arecv(rbuf, 1, FULL)
+                if ep is None:
+                    ep = next(iter(server.list_clients()))
+                await server.asend(ep, sbuf, 2)
+        # Close without flush would drop the last in-flight reply (the
+        # delivery contract the flush tests pin down).
+        await server.aflush()
+        await server.aclose()
+
+    asyncio.run(inner())
+
+
+async def run_sweep(args) -> list[dict]:
+    import starway_amd as sw
+
+    device = args.device
+    results = []
+
+    if args.cross_process:
+        ctx = mp.get_context("spawn")
+        port = 41000 + os.getpid() % 1000
+        p = ctx.Process(target=_server_proc, args=(port, device, SIZES))
+        p.start()
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+
+        async def one_rt(sbuf, rbuf):
+            fut = client.arecv(rbuf, 2, FULL)
+            await client.asend(sbuf, 1)
+            await fut
+
+        for size in SIZES:
+            warmup, iters = pick_iters(size)
+            sbuf = _alloc(size, device, fill=3)
+            rbuf = _alloc(size, device)
+            for _ in range(warmup):
+                await one_rt(sbuf, rbuf)
+            samples = []
+            for _ in range(iters):
+                t0 = time.perf_counter()
+                await one_rt(sbuf, rbuf)
+                samples.append(time.perf_counter() - t0)
+            results.append(_mk_result(size, samples, "cross-process"))
+            print(_fmt(results[-1]), flush=True)
+        await client.aclose()
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+    else:
+        server = sw.Server()
+        client = sw.Client()
+        addr = server.listen_address()
+        await client.aconnect_address(addr)
+        for size in SIZES:
+            warmup, iters = pick_iters(size)
+            sbuf = _alloc(size, device, fill=3)
+            rbuf = _alloc(size, device)
+            pong = _alloc(size, device, fill=1)
+            ep = next(iter(server.list_clients()))
+
+            async def one_rt():
+                sfut = server.arecv(rbuf, 1, FULL)
+                cfut = client.arecv(pong, 2, FULL)
+                await client.asend(sbuf, 1)
+                await sfut
+                await server.asend(ep, sbuf, 2)
+                await cfut
+
+            for _ in range(warmup):
+                await one_rt()
+            samples = []
+            for _ in range(iters):
+                t0 = time.perf_counter()
+                await one_rt()
+                samples.append(time.perf_counter() - t0)
+            results.append(_mk_result(size, samples, "loopback"))
+            print(_fmt(results[-1]), flush=True)
+        await client.aclose()
+        await server.aclose()
+    return results
+
+
+def _mk_result(size: int, samples: list[float], mode: str) -> dict:
+    arr = np.array(samples)
+    p50 = float(np.percentile(arr, 50))
+    return {
+        "size_bytes": size,
+        "mode": mode,
+        "iters": len(samples),
+        "p50_rtt_us": p50 * 1e6,
+        "p50_half_rtt_us": p50 * 1e6 / 2,
+        "gbps_per_direction": size / (p50 / 2) / 1e9,
+    }
+
+
+def _fmt(r: dict) -> str:
+    return (f"{r['size_bytes']:>12} B  p50 half-RTT {r['p50_half_rtt_us']:>10.1f} us  "
+            f"{r['gbps_per_direction']:>8.2f} GB/s/dir  [{r['mode']}]")
+
+
+def main():
+    if os.environ.get("SW_FH"):
+        import faulthandler
+
+        faulthandler.dump_traceback_later(25, exit=True)
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--cross-process", action="store_true")
+    ap.add_argument("--out", default="gpurun_out/sweep.json")
+    args = ap.parse_args()
+    if args.device is None:
+        try:
+            import torch
+
+            args.device = "cuda" if torch.cuda.is_available() else "cpu"
+        except ImportError:
+            args.device = "cpu"
+
+    results = asyncio.run(run_sweep(args))
+    out = Path(args.out)
+    out.parent.mkdir(parents=True, exist_ok=True)
+    out.write_text(json.dumps(
+        {"device": args.device,
+         "cross_process": args.cross_process,
+         "results": results}, indent=2))
+    print(f"wrote {out}")
+
+
+if __name__ == "__main__":
+    main()
