@@ -47,6 +47,9 @@ def parse_args():
                    help="64B pingpong iterations for the latency probe")
     p.add_argument("--device", default=None,
                    help="force 'cpu' or 'cuda' (default: auto)")
+    p.add_argument("--transport", choices=("p2p", "rccl"), default="p2p",
+                   help="p2p: starway tagged engine (hipIpc pulls); "
+                        "rccl: ncclSend/ncclRecv all-to-all groups")
     p.add_argument("--json-out", default=None)
     return p.parse_args()
 
@@ -110,7 +113,33 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
     recv_bufs = {j: alloc(per_peer) for j in peers}
     sync_device()
 
+    mesh = None
+    if args.transport == "rccl":
+        if device == "cpu":
+            raise SystemExit("--transport rccl needs GPUs")
+        import torch
+
+        from starway_amd import rccl as swr
+
+        if world > 1:
+            uid = [swr.unique_id() if rank == 0 else None]
+            dist.broadcast_object_list(uid, src=0)
+            uid = uid[0]
+        else:
+            uid = swr.unique_id()
+        mesh = swr.RcclMesh(uid, rank=rank, world=world,
+                            device=torch.cuda.current_device())
+        # all-to-all layout: peer-major chunks, msg_bytes per rank total.
+        chunk = args.msg_bytes // world
+        rccl_send = alloc(chunk * world, fill=rank % 251)
+        rccl_recv = alloc(chunk * world)
+        sync_device()
+
     async def step(step_idx: int):
+        if mesh is not None:
+            mesh.all_to_all(rccl_send, rccl_recv)
+            mesh.synchronize()
+            return
         recvs = [
             server.arecv(recv_bufs[j], make_tag(j, step_idx), full_mask)
             for j in peers
@@ -234,6 +263,7 @@ def main() -> int:
                 "message_bytes_per_peer": args.msg_bytes // max(1, world - 1),
                 "endpoints": world,
                 "device": device,
+                "transport": args.transport,
                 "parallelism": f"p2p-mesh{world}",
                 "pingpong_64B_half_rtt_us":
                     round(half_rtt_us, 2) if half_rtt_us else None,

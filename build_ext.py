@@ -18,6 +18,7 @@ BUILD = REPO / "build"
 PKG = REPO / "starway_amd"
 
 SOURCES = ["engine.cpp", "gpu.cpp", "module.cpp", "kernels.hip"]
+RCCL_SOURCES = ["rccl_group.cpp"]
 
 HIPCC = os.environ.get("STARWAY_HIPCC", "hipcc")
 ARCH = os.environ.get("STARWAY_OFFLOAD_ARCH", "gfx950")
@@ -81,6 +82,29 @@ def build(verbose: bool = True, debug: bool = False) -> Path:
     out = ext_path()
     if linked_any or not out.exists():
         cmd = [HIPCC, "-shared", "-fPIC", *[str(o) for o in objs], "-o", str(out)]
+        if verbose:
+            print("[build_ext]", " ".join(cmd), flush=True)
+        subprocess.run(cmd, check=True)
+
+    # _rccl: separate module so importing starway_amd does not load librccl.
+    suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+    rccl_out = PKG / f"_rccl{suffix}"
+    rccl_objs: list[Path] = []
+    rccl_linked = False
+    for name in RCCL_SOURCES:
+        src = CSRC / name
+        obj = BUILD / (name.replace(".", "_") + ".o")
+        rccl_objs.append(obj)
+        if _needs_rebuild(obj, src, hdrs):
+            cmd = [HIPCC, *common, "-I", "/opt/rocm/include", "-c", str(src),
+                   "-o", str(obj)]
+            if verbose:
+                print("[build_ext]", " ".join(cmd), flush=True)
+            subprocess.run(cmd, check=True)
+            rccl_linked = True
+    if rccl_linked or not rccl_out.exists():
+        cmd = [HIPCC, "-shared", "-fPIC", *[str(o) for o in rccl_objs],
+               "-L/opt/rocm/lib", "-lrccl", "-o", str(rccl_out)]
         if verbose:
             print("[build_ext]", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True)
