@@ -344,6 +344,9 @@ class Engine {
   std::mutex cmd_mu_;
   std::vector<Op*> cmd_queue_;
   std::atomic<bool> cmd_pending_{false};
+  // True while the engine loop is in hot-spin mode (poll timeout 0):
+  // submitters skip the wake-pipe write, saving a syscall per op.
+  std::atomic<bool> engine_hot_{true};
   int wake_fds_[2] = {-1, -1};  // self-pipe to interrupt poll()
 
   std::thread thread_;

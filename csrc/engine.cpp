@@ -248,6 +248,7 @@ Engine::~Engine() {
 }
 
 void Engine::wake() {
+  if (engine_hot_.load(std::memory_order_acquire)) return;  // already spinning
   char b = 1;
   ssize_t r = ::write(wake_fds_[1], &b, 1);
   (void)r;
@@ -641,7 +642,17 @@ void Engine::loop_iteration(bool& did_work) {
         break;
       }
   if (!busy && idle_iters_ > 50000) timeout = 1;
+  if (timeout != 0) {
+    engine_hot_.store(false, std::memory_order_release);
+    // Re-check for commands that raced the transition (their wake() may
+    // have been skipped while we were still hot).
+    if (cmd_pending_.load(std::memory_order_acquire)) {
+      engine_hot_.store(true, std::memory_order_release);
+      timeout = 0;
+    }
+  }
   poll_sockets(timeout, did_work);
+  engine_hot_.store(true, std::memory_order_release);
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
   if (!pending_flushes_.empty()) check_flush_progress();
   if (!completions_.empty()) {

@@ -52,9 +52,17 @@ int device_of(const void* ptr) {
 // per-device streams
 // ---------------------------------------------------------------------------
 
-static hipStream_t pull_stream(int device) {
-  static std::map<int, hipStream_t> streams;  // guarded by g_mu
-  auto it = streams.find(device);
+// Pulls from DIFFERENT source peers must overlap (each rides its own xGMI
+// link), so streams are keyed by (device, lane) with lane derived from the
+// message source; pulls from the same peer share a lane (one link's worth
+// of bandwidth anyway, and it keeps per-pair completion work ordered).
+constexpr int kStreamLanes = 8;
+
+static hipStream_t pull_stream(int device, int lane = 0) {
+  static std::map<std::pair<int, int>, hipStream_t> streams;  // g_mu held
+  lane &= (kStreamLanes - 1);
+  auto key = std::make_pair(device, lane);
+  auto it = streams.find(key);
   if (it != streams.end()) return it->second;
   int prev;
   hipGetDevice(&prev);
@@ -63,7 +71,7 @@ static hipStream_t pull_stream(int device) {
   if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess)
     s = nullptr;
   hipSetDevice(prev);
-  streams[device] = s;
+  streams[key] = s;
   return s;
 }
 
@@ -150,6 +158,28 @@ struct Ticket {
   RawBuf bounce;  // host staging kept alive until completion
 };
 
+// hipEvent pool per device (create/destroy costs ~1-2 us per op).
+static std::map<int, std::vector<hipEvent_t>> g_event_pool;
+
+static hipError_t pool_get_event(int device, hipEvent_t* ev) {
+  auto& pool = g_event_pool[device];
+  if (!pool.empty()) {
+    *ev = pool.back();
+    pool.pop_back();
+    return hipSuccess;
+  }
+  return hipEventCreateWithFlags(ev, hipEventDisableTiming);
+}
+
+static void pool_put_event(int device, hipEvent_t ev) {
+  auto& pool = g_event_pool[device];
+  if (pool.size() < 256) {
+    pool.push_back(ev);
+  } else {
+    hipEventDestroy(ev);
+  }
+}
+
 static void* resolve_src(const RtsDesc& rts, int open_device,
                          std::string* err) {
   bool same_proc = memcmp(rts.src_uuid, process_uuid(), 16) == 0;
@@ -199,7 +229,10 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   int prev;
   hipGetDevice(&prev);
   hipSetDevice(run_dev);
-  hipStream_t stream = pull_stream(run_dev);
+  // Lane by source peer: device ordinal + a byte of the process uuid so
+  // same-device peers in different processes still spread across lanes.
+  int lane = (rts.device ^ rts.src_uuid[0]) & (kStreamLanes - 1);
+  hipStream_t stream = pull_stream(run_dev, lane);
   hipError_t e;
   if (dst.device >= 0) {
     bool same_proc = memcmp(rts.src_uuid, process_uuid(), 16) == 0;
@@ -215,7 +248,7 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   }
   Ticket* t = new Ticket();
   t->device = run_dev;
-  e = hipEventCreateWithFlags(&t->ev, hipEventDisableTiming);
+  e = pool_get_event(run_dev, &t->ev);
   if (e == hipSuccess) e = hipEventRecord(t->ev, stream);
   hipSetDevice(prev);
   if (e != hipSuccess) {
@@ -243,7 +276,7 @@ void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
   if (e == hipSuccess) {
     t = new Ticket();
     t->device = dst.device;
-    e = hipEventCreateWithFlags(&t->ev, hipEventDisableTiming);
+    e = pool_get_event(dst.device, &t->ev);
     if (e == hipSuccess) e = hipEventRecord(t->ev, stream);
   }
   hipSetDevice(prev);
@@ -270,7 +303,10 @@ int poll_ticket(void* ticket, std::string* err) {
 
 void free_ticket(void* ticket) {
   Ticket* t = (Ticket*)ticket;
-  if (t->ev) hipEventDestroy(t->ev);
+  if (t->ev) {
+    std::lock_guard<std::mutex> lk(g_mu);
+    pool_put_event(t->device, t->ev);
+  }
   delete t;
 }
 
@@ -311,10 +347,12 @@ void synchronize_all() {
   int prev;
   hipGetDevice(&prev);
   for (int d = 0; d < n; d++) {
-    hipStream_t s = pull_stream(d);
-    if (s) {
-      hipSetDevice(d);
-      hipStreamSynchronize(s);
+    for (int lane = 0; lane < kStreamLanes; lane++) {
+      hipStream_t s = pull_stream(d, lane);
+      if (s) {
+        hipSetDevice(d);
+        hipStreamSynchronize(s);
+      }
     }
   }
   hipSetDevice(prev);
