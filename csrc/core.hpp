@@ -91,6 +91,19 @@ struct PeerInfo {
   std::string name;
 };
 
+// Env-tunable knob (config system: STARWAY_* environment variables,
+// the analog of the reference's env-driven configuration surface).
+inline uint64_t env_u64(const char* name, uint64_t dflt) {
+  const char* v = getenv(name);
+  if (!v || !*v) return dflt;
+  char* end = nullptr;
+  unsigned long long x = strtoull(v, &end, 10);
+  if (end && (*end == 'k' || *end == 'K')) x <<= 10;
+  else if (end && (*end == 'm' || *end == 'M')) x <<= 20;
+  else if (end && (*end == 'g' || *end == 'G')) x <<= 30;
+  return (uint64_t)x;
+}
+
 std::vector<uint8_t> encode_peer_info(const PeerInfo& pi);
 bool decode_peer_info(const uint8_t* data, size_t len, PeerInfo* out);
 const uint8_t* process_uuid();  // 16 bytes, stable for this process

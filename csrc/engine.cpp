@@ -60,7 +60,7 @@ static void set_nonblocking(int fd) {
 static void set_tcp_opts(int fd) {
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
-  int buf = 8 * 1024 * 1024;
+  static int buf = (int)env_u64("STARWAY_SOCKBUF", 8 << 20);
   setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
   setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
@@ -460,14 +460,21 @@ double Engine::perf_model(Connection* c, uint64_t msg_size) const {
   // (reference main.cpp:452-467). Constants from xGMI topology (7 links x
   // ~153 GB/s per MI355X) and localhost TCP measurements; recalibrated once
   // rocprof evidence lands (profiles/).
+  // Calibrated against round-1 MI355X measurements (profiles/
+  // r01_sweep_summary.md): device path ~33 us base latency; same-GPU
+  // delivery ~2 TB/s; cross-GPU is xGMI-link-bound (~140 GB/s sustained
+  // per pair). CPU path: localhost TCP ~80 us half-RTT, ~3 GB/s.
   bool peer_gpu = c && c->peer.has_gpu;
   bool self_gpu = gpu::available();
   double lat, bw;
   if (self_gpu && peer_gpu) {
-    lat = 15e-6;           // RTS over TCP + kernel launch + event poll
-    bw = 140e9;            // single xGMI link, sustained
+    bool same_proc = c && memcmp(c->peer.uuid, process_uuid(), 16) == 0;
+    lat = 33e-6;  // RTS over TCP + pull-kernel launch + event poll
+    // Same process usually means same device (loopback benches); across
+    // processes assume the conservative single-link xGMI figure.
+    bw = same_proc ? 2.0e12 : 140e9;
   } else {
-    lat = 25e-6;           // localhost TCP eager
+    lat = 80e-6;  // localhost TCP eager
     bw = 3e9;
   }
   return lat + (double)msg_size / bw;
@@ -641,7 +648,8 @@ void Engine::loop_iteration(bool& did_work) {
         busy = true;
         break;
       }
-  if (!busy && idle_iters_ > 50000) timeout = 1;
+  static const uint64_t kSpinIters = env_u64("STARWAY_SPIN_ITERS", 50000);
+  if (!busy && idle_iters_ > kSpinIters) timeout = 1;
   if (timeout != 0) {
     engine_hot_.store(false, std::memory_order_release);
     // Re-check for commands that raced the transition (their wake() may
@@ -1332,7 +1340,7 @@ void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
 
 void Engine::enqueue_eager(Connection* c, Op* op) {
   TxItem item;
-  constexpr size_t kInline = 4096;
+  static const size_t kInline = env_u64("STARWAY_EAGER_INLINE", 4096);
   FrameHeader h{};
   h.magic = kMagic;
   h.type = FT_EAGER;

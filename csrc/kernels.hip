@@ -117,8 +117,13 @@ hipError_t launch_copy(void* dst, const void* src, size_t bytes,
     size_t n16 = bytes / 16;
     size_t tail = bytes & 15;
     if (n16) {
-      // NT past 64 MiB: the copy would otherwise sweep the 256 MiB LLC.
-      if (bytes >= (64u << 20)) {
+      // NT past 64 MiB (STARWAY_NT_THRESHOLD): the copy would otherwise
+      // sweep the 256 MiB LLC.
+      static const size_t kNtThreshold = [] {
+        const char* v = getenv("STARWAY_NT_THRESHOLD");
+        return v && *v ? (size_t)strtoull(v, nullptr, 10) : (size_t)64 << 20;
+      }();
+      if (bytes >= kNtThreshold) {
         hipLaunchKernelGGL(k_copy_b128_nt, dim3(copy_grid(n16 / 4 + 1)),
                            dim3(256), 0, stream, (const u32x4*)s, (u32x4*)d,
                            n16);
