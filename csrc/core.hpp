@@ -38,7 +38,7 @@ namespace sw {
 // ---------------------------------------------------------------------------
 
 constexpr uint32_t kMagic = 0x53574159;  // "SWAY"
-constexpr uint32_t kProtoVersion = 1;
+constexpr uint32_t kProtoVersion = 2;
 
 enum FrameType : uint16_t {
   FT_HELLO = 1,      // payload: PeerInfo blob (both directions on connect)
@@ -49,6 +49,9 @@ enum FrameType : uint16_t {
   FT_FLUSH_REQ = 6,  // hdr.op_id = flush id; receiver echoes FLUSH_ACK
   FT_FLUSH_ACK = 7,  // hdr.op_id echo
   FT_BYE = 8,        // graceful close notification
+  FT_SHM_OFFER = 10,   // payload: u64 ring cap | shm name — same-host only
+  FT_SHM_ACK = 11,     // flags: 0 accept / 1 decline; sender's LAST tcp frame
+  FT_SHM_SWITCH = 12,  // server's LAST tcp frame after ACK
 };
 
 #pragma pack(push, 1)
@@ -86,6 +89,7 @@ struct RtsDesc {
 struct PeerInfo {
   uint64_t pid = 0;
   uint8_t uuid[16] = {0};
+  uint8_t host_id[16] = {0};  // stable per host (boot id + hostname hash)
   bool has_gpu = false;
   int32_t gpu_count = 0;
   std::string name;
@@ -199,6 +203,7 @@ struct TxItem {
   size_t ext_len = 0;
   bool has_keepalive = false;
   bool is_data = false;           // EAGER/RTS frames: droppable on close
+  bool via_ring = false;          // carried on the shm ring, not the socket
   py::object keepalive;           // dropped (under GIL) once fully written
 };
 
@@ -217,6 +222,8 @@ struct UnexpectedMsg {
   // buffer and the remaining bytes stream straight there (no double copy).
   uint8_t* redirect_dst = nullptr;
 };
+
+class ShmChannel;  // shm.hpp (engine.cpp only)
 
 struct Connection {
   int fd = -1;
@@ -242,6 +249,11 @@ struct Connection {
   bool rx_truncated = false;
   // GPU eager-recv host bounce (posted recv buffer is on device):
   RawBuf rx_gpu_bounce;
+
+  // --- shm channel (same-host fast path) ---
+  std::unique_ptr<ShmChannel> shm;
+  bool shm_tx_enq = false;  // new TxItems go to the ring
+  bool shm_rx = false;      // frames are parsed from the ring
 
   // --- tx ---
   std::deque<TxItem> txq;
@@ -316,6 +328,7 @@ class Engine {
   void do_connect_start();
   void poll_sockets(int timeout_ms, bool& did_work);
   void handle_readable(Connection* c, bool& did_work);
+  void handle_stream(Connection* c, bool from_ring, bool& did_work);
   void handle_writable(Connection* c, bool& did_work);
   void accept_new(bool& did_work);
   void on_frame(Connection* c);           // full header parsed, size==0 or

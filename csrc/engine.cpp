@@ -19,6 +19,7 @@
 //    "cancel" (reference main.cpp:680-701 contract)
 
 #include "core.hpp"
+#include "shm.hpp"
 
 #include <errno.h>
 #include <fcntl.h>
@@ -65,6 +66,34 @@ static void set_tcp_opts(int fd) {
   setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
 
+// Stable per-host identifier: FNV-1a over boot_id + hostname, 16 bytes.
+static const uint8_t* host_id() {
+  static uint8_t id[16] = {0};
+  static bool init = [] {
+    std::string seed;
+    FILE* f = fopen("/proc/sys/kernel/random/boot_id", "r");
+    if (f) {
+      char buf[128];
+      size_t n = fread(buf, 1, sizeof(buf), f);
+      seed.append(buf, n);
+      fclose(f);
+    }
+    char hn[256] = {0};
+    gethostname(hn, sizeof(hn) - 1);
+    seed += hn;
+    uint64_t h1 = 1469598103934665603ull, h2 = 14695981039346656037ull;
+    for (unsigned char ch : seed) {
+      h1 = (h1 ^ ch) * 1099511628211ull;
+      h2 = (h2 ^ (ch + 17)) * 1099511628211ull;
+    }
+    memcpy(id, &h1, 8);
+    memcpy(id + 8, &h2, 8);
+    return true;
+  }();
+  (void)init;
+  return id;
+}
+
 const uint8_t* process_uuid() {
   static uint8_t uuid[16] = {0};
   static bool init = [] {
@@ -90,6 +119,7 @@ std::vector<uint8_t> encode_peer_info(const PeerInfo& pi) {
   put(&ver, 4);
   put(&pi.pid, 8);
   put(pi.uuid, 16);
+  put(pi.host_id, 16);
   uint8_t hg = pi.has_gpu ? 1 : 0;
   put(&hg, 1);
   put(&pi.gpu_count, 4);
@@ -100,7 +130,7 @@ std::vector<uint8_t> encode_peer_info(const PeerInfo& pi) {
 }
 
 bool decode_peer_info(const uint8_t* d, size_t len, PeerInfo* out) {
-  if (len < 4 + 8 + 16 + 1 + 4 + 2) return false;
+  if (len < 4 + 8 + 16 + 16 + 1 + 4 + 2) return false;
   size_t off = 0;
   uint32_t ver;
   memcpy(&ver, d + off, 4);
@@ -109,6 +139,8 @@ bool decode_peer_info(const uint8_t* d, size_t len, PeerInfo* out) {
   memcpy(&out->pid, d + off, 8);
   off += 8;
   memcpy(out->uuid, d + off, 16);
+  off += 16;
+  memcpy(out->host_id, d + off, 16);
   off += 16;
   out->has_gpu = d[off++] != 0;
   memcpy(&out->gpu_count, d + off, 4);
@@ -201,6 +233,7 @@ static PeerInfo self_peer_info(const std::string& name) {
   PeerInfo pi;
   pi.pid = (uint64_t)getpid();
   memcpy(pi.uuid, process_uuid(), 16);
+  memcpy(pi.host_id, host_id(), 16);
   pi.has_gpu = gpu::available();
   pi.gpu_count = gpu::device_count();
   pi.name = name;
@@ -640,11 +673,24 @@ void Engine::loop_iteration(bool& did_work) {
   // Block in poll() only when fully idle; stay hot whenever GPU events or
   // outbound bytes are pending (the ucp_worker_progress spin analog,
   // reference main.cpp:361-468).
+  // Shm rings are polled (no fd): service them every iteration.
+  for (auto& c : conns_) {
+    if (c->dead) continue;
+    if (c->shm_rx && c->shm && c->shm->rx.readable() > 0)
+      handle_stream(c.get(), /*from_ring=*/true, did_work);
+    if (!c->dead && !c->txq.empty() && c->txq.front().via_ring)
+      handle_writable(c.get(), did_work);
+    // Ring EOF: peer closed and everything drained.
+    if (!c->dead && c->shm_rx && c->shm && c->shm->rx.peer_closed() &&
+        c->shm->rx.readable() == 0)
+      on_conn_dead(c.get());
+  }
   int timeout = 0;
   bool busy = !gpu_pulls_.empty();
   if (!busy)
     for (auto& c : conns_)
-      if (c->want_write()) {
+      if (c->want_write() || (c->shm_rx && c->shm && !c->dead &&
+                              c->shm->rx.readable() > 0)) {
         busy = true;
         break;
       }
@@ -827,16 +873,49 @@ void Engine::send_hello(Connection* c) {
 }
 
 void Engine::handle_readable(Connection* c, bool& did_work) {
-  while (!c->dead) {
-    if (c->rx_state == Connection::RxState::Header) {
-      uint8_t* hp = (uint8_t*)&c->rx_hdr;
-      ssize_t n = ::read(c->fd, hp + c->rx_got, sizeof(FrameHeader) - c->rx_got);
+  if (c->shm_rx) {
+    // Post-switch the socket carries no frames: drain it only to detect
+    // peer death (EOF/RST).
+    char scratch[4096];
+    while (true) {
+      ssize_t n = ::read(c->fd, scratch, sizeof(scratch));
       if (n == 0) {
         on_conn_dead(c);
         return;
       }
       if (n < 0) {
         if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+        on_conn_dead(c);
+        return;
+      }
+    }
+  }
+  handle_stream(c, /*from_ring=*/false, did_work);
+}
+
+// Shared frame parser over either byte source. The shm handover switches
+// the source exactly at a frame boundary (SHM_ACK / SHM_SWITCH markers),
+// so one parser state machine serves both.
+void Engine::handle_stream(Connection* c, bool from_ring, bool& did_work) {
+  auto src_read = [&](void* dst, size_t want) -> ssize_t {
+    if (from_ring) {
+      size_t n = c->shm->rx.read(dst, want);
+      if (n == 0) return c->shm->rx.peer_closed() ? 0 : -2;
+      return (ssize_t)n;
+    }
+    ssize_t n = ::read(c->fd, dst, want);
+    if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return -2;
+    return n;
+  };
+  while (!c->dead) {
+    // A mid-stream rx-source switch (SHM_SWITCH parsed from TCP) hands the
+    // remaining frames to the ring reader invoked from the engine loop.
+    if (c->shm_rx != from_ring) return;
+    if (c->rx_state == Connection::RxState::Header) {
+      uint8_t* hp = (uint8_t*)&c->rx_hdr;
+      ssize_t n = src_read(hp + c->rx_got, sizeof(FrameHeader) - c->rx_got);
+      if (n == -2) return;
+      if (n <= 0) {
         on_conn_dead(c);
         return;
       }
@@ -878,13 +957,9 @@ void Engine::handle_readable(Connection* c, bool& did_work) {
         dst = c->rx_small.data() + c->rx_got;
         want = c->rx_hdr.size - c->rx_got;
       }
-      ssize_t n = ::read(c->fd, dst, want);
-      if (n == 0) {
-        on_conn_dead(c);
-        return;
-      }
-      if (n < 0) {
-        if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+      ssize_t n = src_read(dst, want);
+      if (n == -2) return;
+      if (n <= 0) {
         on_conn_dead(c);
         return;
       }
@@ -914,6 +989,7 @@ void Engine::on_frame(Connection* c) {
     case FT_HELLO:
     case FT_RTS:
     case FT_RECV_FAIL:
+    case FT_SHM_OFFER:
       if (h.size > (16 << 20)) {
         on_conn_dead(c);
         return;
@@ -936,6 +1012,23 @@ void Engine::on_frame(Connection* c) {
       return;
     case FT_FLUSH_ACK:
       return;  // legacy; flushes complete on write totals
+    case FT_SHM_ACK:
+      // Client accepted (flags 0) or declined (flags 1) the shm channel.
+      // This was the client's LAST tcp frame: subsequent client frames
+      // arrive on the ring.
+      if (h.flags == 0 && c->shm) {
+        enqueue_frame(c, FT_SHM_SWITCH, 0, 0, 0, nullptr, 0, false);
+        c->shm_tx_enq = true;  // frames after SWITCH ride the ring
+        c->shm_rx = true;
+        if (c->ep) c->ep->transports.emplace_back("sm", "shm_ring");
+      } else {
+        c->shm.reset();
+      }
+      return;
+    case FT_SHM_SWITCH:
+      // Server's LAST tcp frame; its later frames arrive on the ring.
+      c->shm_rx = true;
+      return;
     case FT_BYE:
       on_conn_dead(c);
       return;
@@ -974,6 +1067,43 @@ void Engine::on_frame_payload(Connection* c) {
       on_gpu_send_acked(h.op_id, true, err);
       break;
     }
+    case FT_SHM_OFFER: {
+      // payload: u64 cap | shm name. Client side: map and ACK (accept) or
+      // decline; the ACK is this side's last tcp frame on accept.
+      bool ok = false;
+      const char* shm_env = getenv("STARWAY_SHM");
+      bool shm_enabled = !(shm_env && (!strcmp(shm_env, "0") ||
+                                       !strcmp(shm_env, "false")));
+      if (c->rx_small.size() > 8 && shm_enabled) {
+        uint64_t cap;
+        memcpy(&cap, c->rx_small.data(), 8);
+        std::string name((const char*)c->rx_small.data() + 8,
+                         c->rx_small.size() - 8);
+        std::string err;
+        ShmChannel* ch = ShmChannel::open(name, cap, &err);
+        if (ch) {
+          c->shm.reset(ch);
+          ok = true;
+        } else {
+          SW_DBG("shm open failed: %s", err.c_str());
+        }
+      }
+      {
+        TxItem ack;
+        FrameHeader ah{};
+        ah.magic = kMagic;
+        ah.type = FT_SHM_ACK;
+        ah.flags = ok ? 0 : 1;
+        ack.head.resize(sizeof(ah));
+        memcpy(ack.head.data(), &ah, sizeof(ah));
+        c->tx_enqueued_bytes += ack.head.size();
+        c->txq.push_back(std::move(ack));  // tcp (enqueued pre-switch)
+      }
+      if (ok) c->shm_tx_enq = true;  // everything after the ACK rides the ring
+      bool dummy = false;
+      handle_writable(c, dummy);
+      break;
+    }
     default:
       break;
   }
@@ -1002,6 +1132,35 @@ void Engine::on_hello(Connection* c) {
       eps_.push_back(ep);
     }
     send_hello(c);  // responder reply; unblocks the client's connect
+    // Same-host peer: offer the shared-memory ring channel (disable with
+    // STARWAY_SHM=0 — any value set disables; unset enables).
+    const char* shm_env = getenv("STARWAY_SHM");
+    bool shm_enabled = !(shm_env && (!strcmp(shm_env, "0") ||
+                                     !strcmp(shm_env, "false")));
+    if (memcmp(c->peer.host_id, host_id(), 16) == 0 && shm_enabled) {
+      static std::atomic<uint64_t> shm_seq{1};
+      uint64_t cap = env_u64("STARWAY_SHM_RING", 8 << 20);
+      // round up to a power of two (ring indexing masks with cap-1)
+      uint64_t p2 = 4096;
+      while (p2 < cap) p2 <<= 1;
+      cap = p2;
+      char name[96];
+      snprintf(name, sizeof(name), "/sw-%d-%llu-%llu", (int)getpid(),
+               (unsigned long long)c->conn_id,
+               (unsigned long long)shm_seq.fetch_add(1));
+      std::string err;
+      ShmChannel* ch = ShmChannel::create(name, cap, &err);
+      if (ch) {
+        c->shm.reset(ch);
+        std::vector<uint8_t> payload(8 + strlen(name));
+        memcpy(payload.data(), &cap, 8);
+        memcpy(payload.data() + 8, name, strlen(name));
+        enqueue_frame(c, FT_SHM_OFFER, 0, 0, 0, payload.data(),
+                      payload.size(), false);
+      } else {
+        SW_DBG("shm create failed: %s", err.c_str());
+      }
+    }
     if (have_accept_cb_) {
       Completion comp;
       comp.kind = Completion::Kind::Accept;
@@ -1326,6 +1485,7 @@ void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
   memcpy(item.head.data(), &h, sizeof(h));
   if (payload_len) memcpy(item.head.data() + sizeof(h), payload, payload_len);
   item.is_data = (t == FT_RTS);
+  item.via_ring = c->shm_tx_enq;
   c->tx_enqueued_bytes += item.head.size();
   if (priority && !c->txq.empty()) {
     // Insert at the first frame boundary, after any queued priority frames.
@@ -1349,6 +1509,7 @@ void Engine::enqueue_eager(Connection* c, Op* op) {
   h.op_id = op->id;
   h.aux = op->buf.size;
   item.is_data = true;
+  item.via_ring = c->shm_tx_enq;
   if (op->buf.size <= kInline) {
     item.head.resize(sizeof(h) + op->buf.size);
     memcpy(item.head.data(), &h, sizeof(h));
@@ -1386,6 +1547,17 @@ void Engine::handle_writable(Connection* c, bool& did_work) {
     }
     if (nio == 0) {
       // fully written
+    } else if (it.via_ring) {
+      size_t wrote = 0;
+      for (int k = 0; k < nio; k++) {
+        size_t n = c->shm->tx.write(iov[k].iov_base, iov[k].iov_len);
+        wrote += n;
+        if (n < iov[k].iov_len) break;  // ring full
+      }
+      if (wrote == 0) return;  // consumer will drain; retried by the loop
+      did_work = true;
+      c->tx_front_written += wrote;
+      c->tx_written_bytes += (uint64_t)wrote;
     } else {
       ssize_t n = ::writev(c->fd, iov, nio);
       if (n < 0) {
@@ -1621,8 +1793,14 @@ void Engine::teardown() {
   deadline = std::chrono::steady_clock::now() + std::chrono::milliseconds(200);
   while (std::chrono::steady_clock::now() < deadline) {
     bool any = false;
-    for (auto& c : conns_)
-      if (!c->dead && c->want_write()) any = true;
+    for (auto& c : conns_) {
+      if (c->dead) continue;
+      if (!c->txq.empty() && c->txq.front().via_ring) {
+        bool did = false;
+        handle_writable(c.get(), did);
+      }
+      if (c->want_write()) any = true;
+    }
     if (!any) break;
     bool did = false;
     poll_sockets(5, did);
