@@ -1644,6 +1644,12 @@ void Engine::on_conn_dead(Connection* c) {
   }
   c->txq.clear();
   c->tx_front_written = 0;
+  // Release the shm channel now (marks our tx ring closed so the peer sees
+  // EOF; the creator unlinks the segment name — the peer's own mapping
+  // stays valid until it unmaps).
+  c->shm_rx = false;
+  c->shm_tx_enq = false;
+  c->shm.reset();
   // Flushes whose write target on this conn was not reached can never
   // complete: fail them.
   for (size_t i = 0; i < pending_flushes_.size();) {
@@ -1817,6 +1823,8 @@ void Engine::teardown() {
         item.has_keepalive = false;
       }
     c->txq.clear();
+    c->shm_rx = false;
+    c->shm.reset();
   }
   if (listen_fd_ >= 0) {
     ::close(listen_fd_);
