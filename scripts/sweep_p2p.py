@@ -167,6 +167,69 @@ def _fmt(r: dict) -> str:
             f"{r['gbps_per_direction']:>8.2f} GB/s/dir  [{r['mode']}]")
 
 
+def run_raw_latency(device: str) -> list[dict]:
+    """Raw-callback pingpong (no asyncio): the transport's latency floor.
+    Uses threading.Event for completion wakeups."""
+    import threading
+
+    import starway_amd as sw
+
+    server, client = sw.Server(), sw.Client()
+    addr = server.listen_address()
+    ev = threading.Event()
+    client._client.connect_address(addr, lambda s: ev.set())
+    assert ev.wait(10)
+    ep = None
+    for _ in range(1000):
+        eps = server.list_clients()
+        if eps:
+            ep = next(iter(eps))
+            break
+        time.sleep(0.001)
+
+    results = []
+    for size in [64, 4096, 65536]:
+        ping_out = _alloc(size, device, fill=1)
+        pong_out = _alloc(size, device, fill=2)
+        srv_in = _alloc(size, device)
+        cli_in = _alloc(size, device)
+        warmup, iters = pick_iters(size)
+        samples = []
+
+        def one_rt():
+            # Strict RTT: the server's pong is issued from its ping-recv
+            # completion callback (runs on the server's progress thread).
+            done = threading.Event()
+
+            def on_ping(tag, length):
+                server._server.send(ep, pong_out, 2,
+                                    lambda: None, lambda e: None)
+
+            server._server.recv(srv_in, 1, FULL, on_ping, lambda e: None)
+            client._client.recv(cli_in, 2, FULL,
+                                lambda t, l: done.set(),
+                                lambda e: done.set())
+            client._client.send(ping_out, 1, lambda: None, lambda e: None)
+            assert done.wait(10)
+
+        for _ in range(warmup):
+            one_rt()
+        for _ in range(iters):
+            t0 = time.perf_counter()
+            one_rt()
+            samples.append(time.perf_counter() - t0)
+        results.append(_mk_result(size, samples, "raw-callback"))
+        print(_fmt(results[-1]), flush=True)
+
+    ev2 = threading.Event()
+    client._client.close(lambda: ev2.set())
+    ev2.wait(10)
+    ev3 = threading.Event()
+    server._server.close(lambda: ev3.set())
+    ev3.wait(10)
+    return results
+
+
 def main():
     if os.environ.get("SW_FH"):
         import faulthandler
@@ -175,6 +238,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--device", default=None)
     ap.add_argument("--cross-process", action="store_true")
+    ap.add_argument("--raw", action="store_true",
+                    help="raw-callback latency floor (no asyncio)")
     ap.add_argument("--out", default="gpurun_out/sweep.json")
     args = ap.parse_args()
     if args.device is None:
@@ -185,7 +250,10 @@ def main():
         except ImportError:
             args.device = "cpu"
 
-    results = asyncio.run(run_sweep(args))
+    if args.raw:
+        results = run_raw_latency(args.device)
+    else:
+        results = asyncio.run(run_sweep(args))
     out = Path(args.out)
     out.parent.mkdir(parents=True, exist_ok=True)
     out.write_text(json.dumps(
