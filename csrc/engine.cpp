@@ -572,9 +572,12 @@ void Engine::do_connect_start() {
   if (inet_pton(AF_INET, connect_host_.c_str(), &sin.sin_addr) != 1)
     return fail("invalid address " + connect_host_);
 
-  // Connect with a 5 s budget; ECONNREFUSED is retried (the peer may still
-  // be starting up — common in spawn-subprocess tests and rank rendezvous).
-  auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(5);
+  // Connect with a bounded budget (STARWAY_CONNECT_TIMEOUT seconds,
+  // default 8); ECONNREFUSED is retried within it (the peer may still be
+  // starting up — common in spawn-subprocess tests and rank rendezvous).
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::seconds(
+                      (long)env_u64("STARWAY_CONNECT_TIMEOUT", 8));
   int fd = -1;
   std::string last_err = "connect TIMEOUT";
   while (fd < 0) {

@@ -227,7 +227,7 @@ async def test_gpu_pending_send_canceled_on_close():
 # =============================================================================
 
 
-def _ipc_child_server(port: int, nbytes: int):
+def _ipc_child_server(port: int, nbytes: int, ready):
     import torch
 
     import starway_amd as sw
@@ -238,6 +238,7 @@ def _ipc_child_server(port: int, nbytes: int):
         connected = asyncio.Event()
         loop = asyncio.get_running_loop()
         server.set_accept_cb(lambda ep: loop.call_soon_threadsafe(connected.set))
+        ready.set()  # listening: parent may connect now
         await connected.wait()
         ep = next(iter(server.list_clients()))
         src = torch.arange(nbytes, dtype=torch.uint8, device="cuda") % 251
@@ -253,9 +254,11 @@ def _ipc_child_server(port: int, nbytes: int):
 async def test_cross_process_ipc_device_transfer(port):
     nbytes = 32 << 20
     ctx = mp.get_context("spawn")
-    p = ctx.Process(target=_ipc_child_server, args=(port, nbytes))
+    ready = ctx.Event()
+    p = ctx.Process(target=_ipc_child_server, args=(port, nbytes, ready))
     p.start()
     try:
+        assert ready.wait(120), "child server did not come up"
         client = sw.Client()
         await client.aconnect("127.0.0.1", port)
         dst = torch.zeros(nbytes, dtype=torch.uint8, device="cuda")
