@@ -254,6 +254,7 @@ struct Connection {
   std::unique_ptr<ShmChannel> shm;
   bool shm_tx_enq = false;  // new TxItems go to the ring
   bool shm_rx = false;      // frames are parsed from the ring
+  bool sock_eof = false;    // socket closed; conn dies once the ring drains
 
   // --- tx ---
   std::deque<TxItem> txq;

@@ -683,8 +683,9 @@ void Engine::loop_iteration(bool& did_work) {
       handle_stream(c.get(), /*from_ring=*/true, did_work);
     if (!c->dead && !c->txq.empty() && c->txq.front().via_ring)
       handle_writable(c.get(), did_work);
-    // Ring EOF: peer closed and everything drained.
-    if (!c->dead && c->shm_rx && c->shm && c->shm->rx.peer_closed() &&
+    // Ring EOF: peer closed (ring flag or socket EOF) and ring drained.
+    if (!c->dead && c->shm_rx && c->shm &&
+        (c->shm->rx.peer_closed() || c->sock_eof) &&
         c->shm->rx.readable() == 0)
       on_conn_dead(c.get());
   }
@@ -878,20 +879,21 @@ void Engine::send_hello(Connection* c) {
 void Engine::handle_readable(Connection* c, bool& did_work) {
   if (c->shm_rx) {
     // Post-switch the socket carries no frames: drain it only to detect
-    // peer death (EOF/RST).
+    // peer death (EOF/RST). The peer's closing FIN can arrive BEFORE its
+    // remaining ring frames are consumed — defer connection death until
+    // the ring drains (the engine-loop EOF check below finishes the job).
     char scratch[4096];
-    while (true) {
+    while (c->fd >= 0) {
       ssize_t n = ::read(c->fd, scratch, sizeof(scratch));
-      if (n == 0) {
-        on_conn_dead(c);
-        return;
-      }
-      if (n < 0) {
-        if (errno == EAGAIN || errno == EWOULDBLOCK) return;
-        on_conn_dead(c);
-        return;
-      }
+      if (n > 0) continue;
+      if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
+      // EOF or error: stop polling the socket; ring may still hold data.
+      ::close(c->fd);
+      c->fd = -1;
+      c->sock_eof = true;
+      return;
     }
+    return;
   }
   handle_stream(c, /*from_ring=*/false, did_work);
 }
