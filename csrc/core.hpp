@@ -52,6 +52,7 @@ enum FrameType : uint16_t {
   FT_SHM_OFFER = 10,   // payload: u64 ring cap | shm name — same-host only
   FT_SHM_ACK = 11,     // flags: 0 accept / 1 decline; sender's LAST tcp frame
   FT_SHM_SWITCH = 12,  // server's LAST tcp frame after ACK
+  FT_RTS_CPU = 13,     // payload: CmaDesc — large same-host CPU rendezvous
 };
 
 #pragma pack(push, 1)
@@ -80,6 +81,18 @@ struct RtsDesc {
   uint8_t ipc_handle[kIpcHandleBytes];
   uint64_t offset;   // byte offset of message start within the ipc allocation
   uint64_t raw_ptr;  // device pointer (same-process fast path)
+};
+#pragma pack(pop)
+
+// Same-host CPU rendezvous descriptor (FT_RTS_CPU payload): the receiver
+// pulls straight out of the sender's address space with process_vm_readv
+// (the UCX "cma" transport analog) — one copy, out of band of the byte
+// stream. The sender keeps the buffer pinned until RECV_DONE.
+#pragma pack(push, 1)
+struct CmaDesc {
+  uint64_t pid;       // sender pid for process_vm_readv
+  uint64_t addr;      // source buffer address in the sender
+  uint8_t src_uuid[16];
 };
 #pragma pack(pop)
 
@@ -212,8 +225,10 @@ struct UnexpectedMsg {
   uint64_t size = 0;
   Connection* conn = nullptr;
   bool is_rts = false;
+  bool is_cma = false;
   uint64_t sender_op_id = 0;
   RtsDesc rts{};
+  CmaDesc cma{};
   RawBuf data;  // eager staging (uninitialized alloc)
   size_t got = 0;
   bool complete = false;
@@ -340,6 +355,9 @@ class Engine {
                   uint64_t size, uint64_t sender_op);
   void start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
                       uint64_t size, uint64_t sender_op, Connection* c);
+  void start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
+                      uint64_t size, uint64_t sender_op, Connection* c);
+  void progress_cma(bool& did_work);
   void match_or_stash_recv(Op* op);
   bool try_match_unexpected(Op* op);
   void complete_recv_from_unexpected(Op* op, UnexpectedMsg* um);
@@ -392,6 +410,16 @@ class Engine {
   std::unordered_map<uint64_t, Op*> gpu_sends_;  // op id -> awaiting ack
   struct GpuPull;  // defined in engine.cpp (holds hipEvent)
   std::vector<std::unique_ptr<GpuPull>> gpu_pulls_;
+  struct CmaPull {
+    Op* recv_op = nullptr;
+    Connection* conn = nullptr;
+    uint64_t sender_op_id = 0;
+    uint64_t tag = 0;
+    uint64_t size = 0;
+    uint64_t done = 0;
+    CmaDesc desc{};
+  };
+  std::vector<std::unique_ptr<CmaPull>> cma_pulls_;
   uint64_t next_flush_id_ = 1;
 
   // Connect state (client).

@@ -29,6 +29,7 @@
 #include <poll.h>
 #include <sys/socket.h>
 #include <sys/uio.h>
+#include <sys/syscall.h>
 #include <unistd.h>
 
 #include <chrono>
@@ -690,7 +691,7 @@ void Engine::loop_iteration(bool& did_work) {
       on_conn_dead(c.get());
   }
   int timeout = 0;
-  bool busy = !gpu_pulls_.empty();
+  bool busy = !gpu_pulls_.empty() || !cma_pulls_.empty();
   if (!busy)
     for (auto& c : conns_)
       if (c->want_write() || (c->shm_rx && c->shm && !c->dead &&
@@ -712,6 +713,7 @@ void Engine::loop_iteration(bool& did_work) {
   poll_sockets(timeout, did_work);
   engine_hot_.store(true, std::memory_order_release);
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
+  if (!cma_pulls_.empty()) progress_cma(did_work);
   if (!pending_flushes_.empty()) check_flush_progress();
   if (!completions_.empty()) {
     fire_completions();
@@ -739,6 +741,44 @@ void Engine::process_command(Op* op) {
         return;
       }
       op->conn = c;
+      // Large same-host CPU messages go rendezvous via process_vm_readv
+      // (one copy, out of band). Threshold STARWAY_CMA_THRESHOLD bytes;
+      // STARWAY_CMA=0 disables. Send completes with eager semantics
+      // (handed off) — delivery is still only guaranteed by flush, which
+      // waits for the RECV_DONE of pending CMA ops.
+      static const uint64_t cma_thresh = []() -> uint64_t {
+        const char* v = getenv("STARWAY_CMA");
+        if (v && (!strcmp(v, "0") || !strcmp(v, "false")))
+          return ~0ull;  // disabled
+        return env_u64("STARWAY_CMA_THRESHOLD", 1 << 20);
+      }();
+      if (op->buf.device < 0 && op->buf.size >= cma_thresh &&
+          memcmp(c->peer.host_id, host_id(), 16) == 0 &&
+          memcmp(c->peer.uuid, process_uuid(), 16) != 0) {
+        CmaDesc desc{};
+        desc.pid = (uint64_t)getpid();
+        desc.addr = (uint64_t)(uintptr_t)op->buf.ptr;
+        memcpy(desc.src_uuid, process_uuid(), 16);
+        enqueue_frame(c, FT_RTS_CPU, op->tag, op->id, op->buf.size, &desc,
+                      sizeof(desc), /*priority=*/false);
+        op->gpu_send_awaiting_ack = true;  // same ack machinery as GPU RTS
+        gpu_sends_[op->id] = op;
+        // Eager-style completion: buffer handed off; keepalive stays
+        // pinned in the op until RECV_DONE (or cancel) so the receiver's
+        // pull reads live memory. Both callbacks are consumed here — a
+        // later cancel must not fire fail_cb on an already-completed op.
+        {
+          py::gil_scoped_acquire gil;
+          try {
+            if (op->done_cb.ptr()) op->done_cb();
+          } catch (py::error_already_set& e) {
+            e.discard_as_unraisable("starway send callback");
+          }
+          op->done_cb = py::object();
+          op->fail_cb = py::object();
+        }
+        return;
+      }
       if (op->buf.device >= 0) {
         // GPU rendezvous.
         RtsDesc rts{};
@@ -993,6 +1033,7 @@ void Engine::on_frame(Connection* c) {
       return;
     case FT_HELLO:
     case FT_RTS:
+    case FT_RTS_CPU:
     case FT_RECV_FAIL:
     case FT_SHM_OFFER:
       if (h.size > (16 << 20)) {
@@ -1065,6 +1106,38 @@ void Engine::on_frame_payload(Connection* c) {
       RtsDesc rts;
       memcpy(&rts, c->rx_small.data(), sizeof(rts));
       handle_rts(c, rts, h.tag, h.aux, h.op_id);
+      break;
+    }
+    case FT_RTS_CPU: {
+      if (c->rx_small.size() != sizeof(CmaDesc)) {
+        on_conn_dead(c);
+        return;
+      }
+      CmaDesc cma;
+      memcpy(&cma, c->rx_small.data(), sizeof(cma));
+      // Match like any other message; unmatched waits in the unexpected
+      // queue as a descriptor.
+      bool matched = false;
+      for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
+        Op* r = *it;
+        if ((h.tag & r->tag_mask) == (r->tag & r->tag_mask)) {
+          posted_recvs_.erase(it);
+          start_cma_pull(r, cma, h.tag, h.aux, h.op_id, c);
+          matched = true;
+          break;
+        }
+      }
+      if (!matched) {
+        auto um = std::make_unique<UnexpectedMsg>();
+        um->tag = h.tag;
+        um->size = h.aux;
+        um->conn = c;
+        um->is_cma = true;
+        um->cma = cma;
+        um->sender_op_id = h.op_id;
+        um->complete = true;
+        unexpected_.push_back(std::move(um));
+      }
       break;
     }
     case FT_RECV_FAIL: {
@@ -1330,6 +1403,83 @@ void Engine::start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
   gpu_pulls_.push_back(std::move(pull));
 }
 
+void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
+                            uint64_t size, uint64_t sender_op,
+                            Connection* c) {
+  if (size > recv_op->buf.size) {
+    std::string err = "message truncated (len " + std::to_string(size) +
+                      " > buffer " + std::to_string(recv_op->buf.size) + ")";
+    enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
+                  true);
+    fail_op(recv_op, "receive failed: " + err);
+    return;
+  }
+  auto pull = std::make_unique<CmaPull>();
+  pull->recv_op = recv_op;
+  pull->conn = c;
+  pull->sender_op_id = sender_op;
+  pull->tag = tag;
+  pull->size = size;
+  pull->desc = cma;
+  cma_pulls_.push_back(std::move(pull));
+}
+
+// Chunked process_vm_readv pulls, interleaved with the progress loop so a
+// multi-GiB pull never starves other connections; aborted when the sender
+// dies (its memory is gone — the message is lost, recv re-posts, exactly
+// the unflushed-close contract).
+void Engine::progress_cma(bool& did_work) {
+  constexpr size_t kChunk = 8 << 20;
+  for (size_t i = 0; i < cma_pulls_.size();) {
+    CmaPull* p = cma_pulls_[i].get();
+    if (p->conn && p->conn->dead) {
+      posted_recvs_.push_front(p->recv_op);  // data lost; recv stays pending
+      cma_pulls_.erase(cma_pulls_.begin() + i);
+      continue;
+    }
+    size_t want = (size_t)std::min<uint64_t>(kChunk, p->size - p->done);
+    struct iovec liov {p->recv_op->buf.ptr + p->done, want};
+    struct iovec riov {(void*)(uintptr_t)(p->desc.addr + p->done), want};
+    ssize_t n = process_vm_readv((pid_t)p->desc.pid, &liov, 1, &riov, 1, 0);
+    if (n < 0 && getenv("STARWAY_DEBUG_CMA"))
+      fprintf(stderr, "[sw-cma] readv pid=%llu addr=%llx want=%zu errno=%d (%s)\n",
+              (unsigned long long)p->desc.pid,
+              (unsigned long long)(p->desc.addr + p->done), want, errno,
+              strerror(errno));
+    if (n < 0) {
+      std::string err;
+      if (errno == EPERM || errno == ENOSYS) {
+        err = "cma unavailable (" + std::string(strerror(errno)) + ")";
+      } else {
+        // ESRCH/EFAULT: sender died or freed the buffer mid-pull.
+        posted_recvs_.push_front(p->recv_op);
+        cma_pulls_.erase(cma_pulls_.begin() + i);
+        continue;
+      }
+      enqueue_frame(p->conn, FT_RECV_FAIL, 0, p->sender_op_id, 0, err.data(),
+                    err.size(), true);
+      fail_op(p->recv_op, "receive failed: " + err);
+      cma_pulls_.erase(cma_pulls_.begin() + i);
+      continue;
+    }
+    did_work = true;
+    p->done += (uint64_t)n;
+    if (p->done >= p->size) {
+      enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
+                    true);
+      Completion comp;
+      comp.kind = Completion::Kind::RecvDone;
+      comp.op = p->recv_op;
+      comp.a = p->tag;
+      comp.b = p->size;
+      complete(std::move(comp));
+      cma_pulls_.erase(cma_pulls_.begin() + i);
+      continue;
+    }
+    i++;
+  }
+}
+
 void Engine::poll_gpu(bool& did_work) {
   for (size_t i = 0; i < gpu_pulls_.size();) {
     GpuPull* p = gpu_pulls_[i].get();
@@ -1367,6 +1517,28 @@ void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
   if (it == gpu_sends_.end()) return;
   Op* op = it->second;
   gpu_sends_.erase(it);
+  if (failed && op->buf.device < 0 &&
+      err.find("cma unavailable") != std::string::npos && op->conn &&
+      !op->conn->dead) {
+    // Receiver cannot process_vm_readv us (e.g. yama ptrace restrictions):
+    // retransmit the message as a plain eager stream. The send op already
+    // completed (handed off); extend any pending flush that covered it to
+    // the new write target so flush still means delivery.
+    enqueue_eager(op->conn, op);
+    for (Op* f : pending_flushes_) {
+      if (f->flush_ops_pending.erase(op_id)) {
+        uint64_t target = op->conn->tx_enqueued_bytes;
+        auto [fit, inserted] = f->flush_write_targets.try_emplace(
+            op->conn, target);
+        if (!inserted && fit->second < target) fit->second = target;
+      }
+    }
+    Completion comp;  // silent completion: callbacks already consumed
+    comp.kind = Completion::Kind::SendDone;
+    comp.op = op;
+    complete(std::move(comp));
+    return;
+  }
   if (failed) {
     fail_op(op, "send failed: " + err);
   } else {
@@ -1408,6 +1580,14 @@ bool Engine::try_match_unexpected(Op* op) {
       Connection* c = um->conn;
       unexpected_.erase(it);
       start_gpu_pull(op, rts, tag, size, sop, c);
+      return true;
+    }
+    if (um->is_cma) {
+      CmaDesc cma = um->cma;
+      uint64_t tag = um->tag, size = um->size, sop = um->sender_op_id;
+      Connection* c = um->conn;
+      unexpected_.erase(it);
+      start_cma_pull(op, cma, tag, size, sop, c);
       return true;
     }
     if (!um->complete) {
@@ -1767,6 +1947,9 @@ void Engine::teardown() {
     gpu::free_ticket(p->ticket);
   }
   gpu_pulls_.clear();
+  for (auto& p : cma_pulls_)
+    fail_op(p->recv_op, "operation canceled (endpoint closing)");
+  cma_pulls_.clear();
   // 3. Cancel in-flight data: a connection with undelivered EAGER/RTS bytes
   //    queued is closed abortively — close without flush loses in-flight
   //    sends (the reference's delivery contract, tests/test_basic.py:250-278;

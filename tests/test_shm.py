@@ -117,3 +117,49 @@ async def test_shm_no_segment_leak(port):
     after = set(os.listdir("/dev/shm"))
     leaked = {f for f in after - before if f.startswith("sw-")}
     assert not leaked, f"leaked shm segments: {leaked}"
+
+
+def _cma_child(port, n, env):
+    import os
+
+    os.environ.update(env)
+
+    async def inner():
+        client = Client()
+        await client.aconnect(SERVER_ADDR, port)
+        buf = (np.arange(n) % 247).astype(np.uint8)
+        await client.asend(buf, 31)
+        await client.aflush()
+        await client.aclose()
+
+    asyncio.run(inner())
+
+
+@pytest.mark.parametrize("env", [{}, {"STARWAY_CMA": "0"}])
+async def test_large_cross_process_paths(port, env):
+    """32 MiB cross-process via the CMA rendezvous (default) and via the
+    ring/eager path (STARWAY_CMA=0 in both processes)."""
+    os.environ.update(env)
+    try:
+        n = 32 << 20
+        server = Server()
+        server.listen(SERVER_ADDR, port)
+        ctx = mp.get_context("spawn")
+        p = ctx.Process(target=_cma_child, args=(port, n, env))
+        p.start()
+        try:
+            recv = np.zeros(n, dtype=np.uint8)
+            tag, ln = await server.arecv(recv, 0, 0)
+            assert tag == 31 and ln == n
+            expect = (np.arange(n) % 247).astype(np.uint8)
+            np.testing.assert_array_equal(recv, expect)
+        finally:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.kill()
+                p.join()
+            p.close()
+            await server.aclose()
+    finally:
+        for k in env:
+            os.environ.pop(k, None)
