@@ -420,7 +420,6 @@ class Engine {
     CmaDesc desc{};
   };
   std::vector<std::unique_ptr<CmaPull>> cma_pulls_;
-  uint64_t next_flush_id_ = 1;
 
   // Connect state (client).
   std::string connect_host_;
