@@ -424,6 +424,8 @@ class Engine {
   // Connect state (client).
   std::string connect_host_;
   int connect_port_ = 0;
+  // Worker-address mode: fallback routes tried in order.
+  std::vector<std::pair<std::string, int>> connect_candidates_;
   py::object connect_cb_;
   bool connect_requested_ = false;
   bool connect_done_ = false;

@@ -376,12 +376,12 @@ void Engine::connect_address(const std::vector<uint8_t>& blob, py::object cb) {
   PeerInfo pi;
   if (!decode_worker_address(blob, &addrs, &pi) || addrs.empty())
     throw std::runtime_error("invalid worker address blob");
-  // Try the first address; remaining candidates joined for the engine to try.
   int expect = 0;
   if (!status_.compare_exchange_strong(expect, 1))
     throw std::runtime_error("client already connected or closed");
   connect_host_ = addrs[0].first;
   connect_port_ = addrs[0].second;
+  connect_candidates_ = std::move(addrs);
   connect_cb_ = std::move(cb);
   connect_requested_ = true;
   thread_ = std::thread([this] { thread_main(); });
@@ -567,11 +567,11 @@ void Engine::do_connect_start() {
     status_.store(4, std::memory_order_release);
   };
 
-  struct sockaddr_in sin {};
-  sin.sin_family = AF_INET;
-  sin.sin_port = htons((uint16_t)connect_port_);
-  if (inet_pton(AF_INET, connect_host_.c_str(), &sin.sin_addr) != 1)
-    return fail("invalid address " + connect_host_);
+  // Route candidates: worker-address mode may advertise several host IPs;
+  // they are tried round-robin until one connects.
+  if (connect_candidates_.empty())
+    connect_candidates_.emplace_back(connect_host_, connect_port_);
+  size_t cand_idx = 0;
 
   // Connect with a bounded budget (STARWAY_CONNECT_TIMEOUT seconds,
   // default 8); ECONNREFUSED is retried within it (the peer may still be
@@ -584,6 +584,17 @@ void Engine::do_connect_start() {
   while (fd < 0) {
     if (status_.load(std::memory_order_acquire) == 3) return fail("canceled");
     if (std::chrono::steady_clock::now() > deadline) return fail(last_err);
+    auto& [host, port] = connect_candidates_[cand_idx % connect_candidates_.size()];
+    cand_idx++;
+    connect_host_ = host;
+    connect_port_ = port;
+    struct sockaddr_in sin {};
+    sin.sin_family = AF_INET;
+    sin.sin_port = htons((uint16_t)port);
+    if (inet_pton(AF_INET, host.c_str(), &sin.sin_addr) != 1) {
+      last_err = "invalid address " + host;
+      continue;
+    }
     int s = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
     if (s < 0) return fail("socket() failed");
     set_nonblocking(s);
@@ -621,9 +632,12 @@ void Engine::do_connect_start() {
     }
     ::close(s);
     last_err = std::string("connect: ") + strerror(cerr);
-    if (cerr != ECONNREFUSED && cerr != ENETUNREACH && cerr != EHOSTUNREACH)
+    if (cerr != ECONNREFUSED && cerr != ENETUNREACH && cerr != EHOSTUNREACH &&
+        cerr != ETIMEDOUT)
       return fail(last_err);
-    usleep(20000);  // refused: server not up yet; retry until deadline
+    // refused/unreachable: rotate to the next candidate; back off only
+    // once all candidates were tried this round.
+    if (cand_idx % connect_candidates_.size() == 0) usleep(20000);
   }
 
   auto c = std::make_unique<Connection>();

@@ -23,6 +23,9 @@ class HipccBuildExt(_build_ext):
 
 
 setup(
+    name="starway-amd",
+    version="0.1.0",
+    packages=["starway_amd", "starway_amd.benchmarks"],
     cmdclass={"build_ext": HipccBuildExt},
     # A dummy ext module so build_ext runs under `pip install`/wheel builds.
     ext_modules=[],
