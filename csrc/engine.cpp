@@ -1454,25 +1454,36 @@ void Engine::progress_cma(bool& did_work) {
     size_t want = (size_t)std::min<uint64_t>(kChunk, p->size - p->done);
     struct iovec liov {p->recv_op->buf.ptr + p->done, want};
     struct iovec riov {(void*)(uintptr_t)(p->desc.addr + p->done), want};
-    ssize_t n = process_vm_readv((pid_t)p->desc.pid, &liov, 1, &riov, 1, 0);
+    ssize_t n;
+    static const bool force_eperm =
+        getenv("STARWAY_CMA_FORCE_EPERM") != nullptr;  // test hook
+    if (force_eperm) {
+      errno = EPERM;
+      n = -1;
+    } else {
+      n = process_vm_readv((pid_t)p->desc.pid, &liov, 1, &riov, 1, 0);
+    }
     if (n < 0 && getenv("STARWAY_DEBUG_CMA"))
       fprintf(stderr, "[sw-cma] readv pid=%llu addr=%llx want=%zu errno=%d (%s)\n",
               (unsigned long long)p->desc.pid,
               (unsigned long long)(p->desc.addr + p->done), want, errno,
               strerror(errno));
     if (n < 0) {
-      std::string err;
       if (errno == EPERM || errno == ENOSYS) {
-        err = "cma unavailable (" + std::string(strerror(errno)) + ")";
-      } else {
-        // ESRCH/EFAULT: sender died or freed the buffer mid-pull.
+        // No ptrace rights to the sender (e.g. yama scope: a child cannot
+        // read its parent). Tell the sender to retransmit as eager and
+        // RE-POST the recv so the retransmission matches it.
+        std::string err =
+            "cma unavailable (" + std::string(strerror(errno)) + ")";
+        enqueue_frame(p->conn, FT_RECV_FAIL, 0, p->sender_op_id, 0,
+                      err.data(), err.size(), true);
         posted_recvs_.push_front(p->recv_op);
-        cma_pulls_.erase(cma_pulls_.begin() + i);
-        continue;
+      } else {
+        // ESRCH/EFAULT: sender died or freed the buffer mid-pull — the
+        // message is lost; the recv stays pending (unflushed-close
+        // contract).
+        posted_recvs_.push_front(p->recv_op);
       }
-      enqueue_frame(p->conn, FT_RECV_FAIL, 0, p->sender_op_id, 0, err.data(),
-                    err.size(), true);
-      fail_op(p->recv_op, "receive failed: " + err);
       cma_pulls_.erase(cma_pulls_.begin() + i);
       continue;
     }

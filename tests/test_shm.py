@@ -135,7 +135,11 @@ def _cma_child(port, n, env):
     asyncio.run(inner())
 
 
-@pytest.mark.parametrize("env", [{}, {"STARWAY_CMA": "0"}])
+@pytest.mark.parametrize(
+    "env",
+    [{}, {"STARWAY_CMA": "0"}, {"STARWAY_CMA_FORCE_EPERM": "1"}],
+    ids=["cma", "ring-eager", "cma-eperm-fallback"],
+)
 async def test_large_cross_process_paths(port, env):
     """32 MiB cross-process via the CMA rendezvous (default) and via the
     ring/eager path (STARWAY_CMA=0 in both processes)."""
