@@ -270,6 +270,7 @@ struct Connection {
   bool shm_tx_enq = false;  // new TxItems go to the ring
   bool shm_rx = false;      // frames are parsed from the ring
   bool sock_eof = false;    // socket closed; conn dies once the ring drains
+  bool cma_denied = false;  // peer reported 'cma unavailable': send eager
 
   // --- tx ---
   std::deque<TxItem> txq;

@@ -767,6 +767,7 @@ void Engine::process_command(Op* op) {
         return env_u64("STARWAY_CMA_THRESHOLD", 1 << 20);
       }();
       if (op->buf.device < 0 && op->buf.size >= cma_thresh &&
+          !c->cma_denied &&
           memcmp(c->peer.host_id, host_id(), 16) == 0 &&
           memcmp(c->peer.uuid, process_uuid(), 16) != 0) {
         CmaDesc desc{};
@@ -1545,6 +1546,7 @@ void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
   if (failed && op->buf.device < 0 &&
       err.find("cma unavailable") != std::string::npos && op->conn &&
       !op->conn->dead) {
+    op->conn->cma_denied = true;  // stop offering CMA on this connection
     // Receiver cannot process_vm_readv us (e.g. yama ptrace restrictions):
     // retransmit the message as a plain eager stream. The send op already
     // completed (handed off); extend any pending flush that covered it to
