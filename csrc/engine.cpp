@@ -1685,6 +1685,7 @@ void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
 void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
                            uint64_t op_id, uint64_t aux, const void* payload,
                            size_t payload_len, bool priority) {
+  if (c->dead) return;
   TxItem item;
   item.head.resize(sizeof(FrameHeader) + payload_len);
   FrameHeader h{};
