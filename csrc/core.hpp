@@ -336,6 +336,16 @@ class Engine {
 
   int status() const { return status_.load(std::memory_order_acquire); }
 
+  // Observability counters (engine thread writes; reads are racy-but-
+  // monotonic snapshots, fine for stats).
+  struct Stats {
+    std::atomic<uint64_t> msgs_sent{0}, msgs_received{0};
+    std::atomic<uint64_t> bytes_sent{0}, bytes_received{0};
+    std::atomic<uint64_t> eager_rx{0}, gpu_rx{0}, cma_rx{0};
+    std::atomic<uint64_t> unexpected_rx{0};
+  };
+  Stats stats_;
+
  private:
   // ---- engine thread ----
   void thread_main();

@@ -755,6 +755,8 @@ void Engine::process_command(Op* op) {
         return;
       }
       op->conn = c;
+      stats_.msgs_sent.fetch_add(1, std::memory_order_relaxed);
+      stats_.bytes_sent.fetch_add(op->buf.size, std::memory_order_relaxed);
       // Large same-host CPU messages go rendezvous via process_vm_readv
       // (one copy, out of band). Threshold STARWAY_CMA_THRESHOLD bytes;
       // STARWAY_CMA=0 disables. Send completes with eager semantics
@@ -1292,6 +1294,7 @@ void Engine::begin_eager(Connection* c) {
     }
   }
   if (!c->rx_recv_op) {
+    stats_.unexpected_rx.fetch_add(1, std::memory_order_relaxed);
     auto um = std::make_unique<UnexpectedMsg>();
     um->tag = tag;
     um->size = msg_len;
@@ -1341,6 +1344,9 @@ void Engine::finish_eager_into_recv(Connection* c) {
       gpu_pulls_.push_back(std::move(pull));
       return;
     }
+    stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+    stats_.bytes_received.fetch_add(r->recv_len, std::memory_order_relaxed);
+    stats_.eager_rx.fetch_add(1, std::memory_order_relaxed);
     Completion comp;
     comp.kind = Completion::Kind::RecvDone;
     comp.op = r;
@@ -1493,6 +1499,9 @@ void Engine::progress_cma(bool& did_work) {
     if (p->done >= p->size) {
       enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
                     true);
+      stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+      stats_.bytes_received.fetch_add(p->size, std::memory_order_relaxed);
+      stats_.cma_rx.fetch_add(1, std::memory_order_relaxed);
       Completion comp;
       comp.kind = Completion::Kind::RecvDone;
       comp.op = p->recv_op;
@@ -1520,6 +1529,9 @@ void Engine::poll_gpu(bool& did_work) {
       if (p->sender_op_id && p->conn && !p->conn->dead)
         enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
                       true);
+      stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+      stats_.bytes_received.fetch_add(p->len, std::memory_order_relaxed);
+      stats_.gpu_rx.fetch_add(1, std::memory_order_relaxed);
       Completion comp;
       comp.kind = Completion::Kind::RecvDone;
       comp.op = p->recv_op;
@@ -1672,6 +1684,9 @@ void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
     return;
   }
   memcpy(op->buf.ptr, um->data.data(), um->size);
+  stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+  stats_.bytes_received.fetch_add(um->size, std::memory_order_relaxed);
+  stats_.eager_rx.fetch_add(1, std::memory_order_relaxed);
   Completion comp;
   comp.kind = Completion::Kind::RecvDone;
   comp.op = op;

@@ -162,6 +162,20 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("client_ep"), py::arg("done_callback"),
            py::arg("fail_callback"))
+      .def("get_stats",
+           [](PyServer& s) {
+             py::dict d;
+             auto& st = s.engine.stats_;
+             d["msgs_sent"] = st.msgs_sent.load();
+             d["msgs_received"] = st.msgs_received.load();
+             d["bytes_sent"] = st.bytes_sent.load();
+             d["bytes_received"] = st.bytes_received.load();
+             d["eager_rx"] = st.eager_rx.load();
+             d["gpu_rx"] = st.gpu_rx.load();
+             d["cma_rx"] = st.cma_rx.load();
+             d["unexpected_rx"] = st.unexpected_rx.load();
+             return d;
+           })
       .def("list_clients",
            [](PyServer& s) {
              py::set out;
@@ -220,6 +234,20 @@ PYBIND11_MODULE(_core, m) {
              c.engine.flush(std::move(done), std::move(fail));
            },
            py::arg("done_callback"), py::arg("fail_callback"))
+      .def("get_stats",
+           [](PyClient& c) {
+             py::dict d;
+             auto& st = c.engine.stats_;
+             d["msgs_sent"] = st.msgs_sent.load();
+             d["msgs_received"] = st.msgs_received.load();
+             d["bytes_sent"] = st.bytes_sent.load();
+             d["bytes_received"] = st.bytes_received.load();
+             d["eager_rx"] = st.eager_rx.load();
+             d["gpu_rx"] = st.gpu_rx.load();
+             d["cma_rx"] = st.cma_rx.load();
+             d["unexpected_rx"] = st.unexpected_rx.load();
+             return d;
+           })
       .def("evaluate_perf", [](PyClient& c, uint64_t msg_size) {
         return c.engine.evaluate_perf(nullptr, msg_size);
       },

@@ -245,3 +245,19 @@ def test_bench_cli_loopback_all_scenarios(tmp_path, port):
     assert by_name["small-messages"]["metrics"]["messages_per_second"] > 0
     assert by_name["pingpong-flag"]["metrics"]["avg_rtt_us"] > 0
     assert by_name["streaming-duplex"]["metrics"]["aggregate_gbps"] > 0
+
+
+async def test_stats_counters(port):
+    async with pair(port) as (server, client):
+        msg = np.arange(64, dtype=np.uint8)
+        buf = np.zeros(64, dtype=np.uint8)
+        for i in range(3):
+            fut = server.arecv(buf, 0, 0)
+            await asyncio.sleep(0.005)
+            await client.asend(msg, i)
+            await fut
+        cs = client._client.get_stats()
+        ss = server._server.get_stats()
+        assert cs["msgs_sent"] == 3 and cs["bytes_sent"] == 192
+        assert ss["msgs_received"] == 3 and ss["bytes_received"] == 192
+        assert ss["eager_rx"] == 3
