@@ -108,6 +108,19 @@ def build(verbose: bool = True, debug: bool = False) -> Path:
         if verbose:
             print("[build_ext]", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True)
+
+    # bin/copy_bench: standalone kernel microbench (PMC profiling target).
+    bench_bin = REPO / "bin" / "copy_bench"
+    bench_src = CSRC / "copy_bench_main.hip"
+    bench_obj = BUILD / "copy_bench_main.o"
+    if bench_src.exists():
+        (REPO / "bin").mkdir(exist_ok=True)
+        if _needs_rebuild(bench_obj, bench_src, hdrs) or not bench_bin.exists():
+            subprocess.run([HIPCC, *common, "-c", str(bench_src), "-o",
+                            str(bench_obj)], check=True)
+            subprocess.run([HIPCC, str(bench_obj),
+                            str(BUILD / "kernels_hip.o"), "-o",
+                            str(bench_bin)], check=True)
     return out
 
 
