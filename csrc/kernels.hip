@@ -71,6 +71,36 @@ __global__ __launch_bounds__(256) void k_copy_b128_nt(
   }
 }
 
+// Deeper-unroll NT variant (8 independent 16B elements in flight/thread).
+__global__ __launch_bounds__(256) void k_copy_b128_nt_u8(
+    const u32x4* __restrict__ src, u32x4* __restrict__ dst, size_t n16) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  while (i + 7 * stride < n16) {
+    u32x4 v0 = __builtin_nontemporal_load(&src[i]);
+    u32x4 v1 = __builtin_nontemporal_load(&src[i + stride]);
+    u32x4 v2 = __builtin_nontemporal_load(&src[i + 2 * stride]);
+    u32x4 v3 = __builtin_nontemporal_load(&src[i + 3 * stride]);
+    u32x4 v4 = __builtin_nontemporal_load(&src[i + 4 * stride]);
+    u32x4 v5 = __builtin_nontemporal_load(&src[i + 5 * stride]);
+    u32x4 v6 = __builtin_nontemporal_load(&src[i + 6 * stride]);
+    u32x4 v7 = __builtin_nontemporal_load(&src[i + 7 * stride]);
+    __builtin_nontemporal_store(v0, &dst[i]);
+    __builtin_nontemporal_store(v1, &dst[i + stride]);
+    __builtin_nontemporal_store(v2, &dst[i + 2 * stride]);
+    __builtin_nontemporal_store(v3, &dst[i + 3 * stride]);
+    __builtin_nontemporal_store(v4, &dst[i + 4 * stride]);
+    __builtin_nontemporal_store(v5, &dst[i + 5 * stride]);
+    __builtin_nontemporal_store(v6, &dst[i + 6 * stride]);
+    __builtin_nontemporal_store(v7, &dst[i + 7 * stride]);
+    i += 8 * stride;
+  }
+  for (; i < n16; i += stride) {
+    u32x4 a = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(a, &dst[i]);
+  }
+}
+
 // Byte-granularity fallback for arbitrary (mis)alignment.
 __global__ __launch_bounds__(256) void k_copy_b8(
     const uint8_t* __restrict__ src, uint8_t* __restrict__ dst, size_t n) {
@@ -93,9 +123,14 @@ __global__ __launch_bounds__(256) void k_copy_b32(
 
 static inline int copy_grid(size_t work_items) {
   // >> 256 WGs to fill 8 XCDs x 32 CUs; cap so tiny copies stay one-wave-ish.
+  // Cap tunable via STARWAY_COPY_BLOCKS for on-box sweeps.
+  static const size_t cap = [] {
+    const char* v = getenv("STARWAY_COPY_BLOCKS");
+    return v && *v ? strtoull(v, nullptr, 10) : 8192ull;
+  }();
   size_t blocks = (work_items + 255) / 256;
   if (blocks < 1) blocks = 1;
-  if (blocks > 8192) blocks = 8192;
+  if (blocks > cap) blocks = cap;
   return (int)blocks;
 }
 
@@ -123,10 +158,20 @@ hipError_t launch_copy(void* dst, const void* src, size_t bytes,
         const char* v = getenv("STARWAY_NT_THRESHOLD");
         return v && *v ? (size_t)strtoull(v, nullptr, 10) : (size_t)64 << 20;
       }();
+      static const int nt_unroll = [] {
+        const char* v = getenv("STARWAY_COPY_UNROLL");
+        return v && *v ? atoi(v) : 4;
+      }();
       if (bytes >= kNtThreshold) {
-        hipLaunchKernelGGL(k_copy_b128_nt, dim3(copy_grid(n16 / 4 + 1)),
-                           dim3(256), 0, stream, (const u32x4*)s, (u32x4*)d,
-                           n16);
+        if (nt_unroll >= 8) {
+          hipLaunchKernelGGL(k_copy_b128_nt_u8, dim3(copy_grid(n16 / 8 + 1)),
+                             dim3(256), 0, stream, (const u32x4*)s,
+                             (u32x4*)d, n16);
+        } else {
+          hipLaunchKernelGGL(k_copy_b128_nt, dim3(copy_grid(n16 / 4 + 1)),
+                             dim3(256), 0, stream, (const u32x4*)s,
+                             (u32x4*)d, n16);
+        }
       } else {
         hipLaunchKernelGGL(k_copy_b128, dim3(copy_grid(n16 / 4 + 1)),
                            dim3(256), 0, stream, (const uint4*)s, (uint4*)d,
