@@ -69,6 +69,10 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--device", default=None,
                    help="Buffer device for all scenarios: cpu (default) or "
                         "cuda / cuda:N.")
+    p.add_argument("--tls",
+                   help="Accepted for reference-CLI compatibility; the "
+                        "native transport selects tcp/shm/cma/xgmi "
+                        "automatically (use STARWAY_* env knobs to pin).")
     p.add_argument("--scenarios", nargs="*",
                    help="Scenarios (default all): " + ", ".join(list_scenarios()))
     p.add_argument("--large-bytes", type=parse_size)
@@ -319,6 +323,9 @@ def dump_results(results: Sequence[ScenarioResult], args: argparse.Namespace) ->
 
 def main(argv: Sequence[str] | None = None) -> int:
     args = build_parser().parse_args(argv)
+    if args.tls:
+        print(f"[bench] --tls {args.tls} noted; transports are selected "
+              "natively (tcp/shm_ring/cma/xgmi) — see STARWAY_* env knobs.")
     if args.role == "server":
         asyncio.run(run_server(args))
         return 0
