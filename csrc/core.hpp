@@ -81,6 +81,11 @@ struct RtsDesc {
   uint8_t ipc_handle[kIpcHandleBytes];
   uint64_t offset;   // byte offset of message start within the ipc allocation
   uint64_t raw_ptr;  // device pointer (same-process fast path)
+  // Source layout (0 rows => contiguous): pack happens inside the pull
+  // kernel, so non-contiguous tensors move without a .contiguous() pass.
+  uint64_t src_rows;
+  uint64_t src_row_bytes;
+  uint64_t src_stride;
 };
 #pragma pack(pop)
 
@@ -130,10 +135,16 @@ const uint8_t* process_uuid();  // 16 bytes, stable for this process
 // ---------------------------------------------------------------------------
 
 // A caller-provided message buffer. device < 0 => host memory.
+// rows > 0 => 2D-strided device buffer (non-contiguous tensor): the
+// message is rows x row_bytes dense bytes, row r starting at
+// ptr + r*stride; size == rows*row_bytes.
 struct BufferRef {
   uint8_t* ptr = nullptr;
   size_t size = 0;
   int device = -1;
+  uint64_t rows = 0;
+  uint64_t row_bytes = 0;
+  uint64_t stride = 0;
 };
 
 // Uninitialized heap buffer (std::vector zero-fills on resize, which stalls

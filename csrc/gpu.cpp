@@ -18,6 +18,10 @@ namespace sw {
 // kernels.hip
 hipError_t launch_copy(void* dst, const void* src, size_t bytes,
                        hipStream_t stream);
+hipError_t launch_copy_strided(void* dst, uint64_t dst_stride,
+                               const void* src, uint64_t src_stride,
+                               uint64_t rows, uint64_t row_bytes,
+                               hipStream_t stream);
 
 namespace gpu {
 
@@ -118,6 +122,9 @@ bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err) {
   memcpy(out->src_uuid, process_uuid(), 16);
   out->device = buf.device;
   out->raw_ptr = (uint64_t)(uintptr_t)buf.ptr;
+  out->src_rows = buf.rows;
+  out->src_row_bytes = buf.row_bytes;
+  out->src_stride = buf.stride;
   // Resolve the allocation base for IPC export.
   int prev;
   hipGetDevice(&prev);
@@ -245,7 +252,31 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
       const char* v = getenv("STARWAY_PULL");
       return v && strcmp(v, "sdma") == 0;
     }();
-    if (use_sdma) {
+    bool src_strided = rts.src_rows > 0;
+    bool dst_strided = dst.rows > 0;
+    if (dst_strided && dst.size != size) {
+      hipSetDevice(prev);
+      *err = "strided recv buffer geometry mismatch (buffer " +
+             std::to_string(dst.size) + " B, message " +
+             std::to_string(size) + " B)";
+      return nullptr;
+    }
+    if (src_strided || dst_strided) {
+      // Pack/unpack inside the pull kernel. Row geometry: prefer the
+      // source's; a strided dst must agree on row_bytes.
+      uint64_t rows = src_strided ? rts.src_rows : dst.rows;
+      uint64_t row_bytes = src_strided ? rts.src_row_bytes : dst.row_bytes;
+      if (dst_strided && src_strided &&
+          (dst.rows != rows || dst.row_bytes != row_bytes)) {
+        hipSetDevice(prev);
+        *err = "strided send/recv row geometry mismatch";
+        return nullptr;
+      }
+      uint64_t sstride = src_strided ? rts.src_stride : row_bytes;
+      uint64_t dstride = dst_strided ? dst.stride : row_bytes;
+      e = launch_copy_strided(dst.ptr, dstride, src, sstride, rows,
+                              row_bytes, stream);
+    } else if (use_sdma) {
       if (same_proc && rts.device != dst.device) {
         e = hipMemcpyPeerAsync(dst.ptr, dst.device, src, rts.device, size,
                                stream);
@@ -257,7 +288,13 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
       e = launch_copy(dst.ptr, src, size, stream);
     }
   } else {
-    e = hipMemcpyAsync(dst.ptr, src, size, hipMemcpyDeviceToHost, stream);
+    if (rts.src_rows > 0) {
+      e = hipMemcpy2DAsync(dst.ptr, rts.src_row_bytes, src, rts.src_stride,
+                           rts.src_row_bytes, rts.src_rows,
+                           hipMemcpyDeviceToHost, stream);
+    } else {
+      e = hipMemcpyAsync(dst.ptr, src, size, hipMemcpyDeviceToHost, stream);
+    }
   }
   if (e != hipSuccess) {
     hipSetDevice(prev);
@@ -288,8 +325,14 @@ void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
   hipGetDevice(&prev);
   hipSetDevice(dst.device);
   hipStream_t stream = pull_stream(dst.device);
-  hipError_t e =
-      hipMemcpyAsync(dst.ptr, src, size, hipMemcpyHostToDevice, stream);
+  hipError_t e;
+  if (dst.rows > 0) {
+    e = hipMemcpy2DAsync(dst.ptr, dst.stride, src, dst.row_bytes,
+                         dst.row_bytes, dst.rows, hipMemcpyHostToDevice,
+                         stream);
+  } else {
+    e = hipMemcpyAsync(dst.ptr, src, size, hipMemcpyHostToDevice, stream);
+  }
   Ticket* t = nullptr;
   if (e == hipSuccess) {
     t = new Ticket();

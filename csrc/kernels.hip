@@ -118,6 +118,43 @@ __global__ __launch_bounds__(256) void k_copy_b32(
 }
 
 // ---------------------------------------------------------------------------
+// Strided (2D) pack/unpack copy: row r of the message lives at
+// src + r*src_stride and lands at dst + r*dst_stride; rows are dense
+// row_bytes-long runs. Used for non-contiguous device tensors (e.g. torch
+// slices) so they move without a .contiguous() staging pass. Lanes walk
+// 16B elements in row-major message order, so global accesses stay
+// coalesced within each row.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_copy_strided_b128(
+    const uint8_t* __restrict__ src, uint64_t src_stride,
+    uint8_t* __restrict__ dst, uint64_t dst_stride, uint64_t rows,
+    uint64_t row_n16) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t total = (size_t)rows * row_n16;
+  for (; i < total; i += stride) {
+    size_t r = i / row_n16;
+    size_t c = i - r * row_n16;
+    *(uint4*)(dst + r * dst_stride + c * 16) =
+        *(const uint4*)(src + r * src_stride + c * 16);
+  }
+}
+
+__global__ __launch_bounds__(256) void k_copy_strided_b8(
+    const uint8_t* __restrict__ src, uint64_t src_stride,
+    uint8_t* __restrict__ dst, uint64_t dst_stride, uint64_t rows,
+    uint64_t row_bytes) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t total = (size_t)rows * row_bytes;
+  for (; i < total; i += stride) {
+    size_t r = i / row_bytes;
+    size_t c = i - r * row_bytes;
+    dst[r * dst_stride + c] = src[r * src_stride + c];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Host-side dispatch
 // ---------------------------------------------------------------------------
 
@@ -190,6 +227,29 @@ hipError_t launch_copy(void* dst, const void* src, size_t bytes,
   } else {
     hipLaunchKernelGGL(k_copy_b8, dim3(copy_grid(bytes / 16 + 1)), dim3(256),
                        0, stream, (const uint8_t*)src, (uint8_t*)dst, bytes);
+  }
+  return hipGetLastError();
+}
+
+hipError_t launch_copy_strided(void* dst, uint64_t dst_stride,
+                               const void* src, uint64_t src_stride,
+                               uint64_t rows, uint64_t row_bytes,
+                               hipStream_t stream) {
+  if (rows == 0 || row_bytes == 0) return hipSuccess;
+  uintptr_t s = (uintptr_t)src, d = (uintptr_t)dst;
+  bool vec = (s & 15) == 0 && (d & 15) == 0 && (src_stride & 15) == 0 &&
+             (dst_stride & 15) == 0 && (row_bytes & 15) == 0;
+  if (vec) {
+    uint64_t row_n16 = row_bytes / 16;
+    size_t total = (size_t)rows * row_n16;
+    hipLaunchKernelGGL(k_copy_strided_b128, dim3(copy_grid(total / 4 + 1)),
+                       dim3(256), 0, stream, (const uint8_t*)src, src_stride,
+                       (uint8_t*)dst, dst_stride, rows, row_n16);
+  } else {
+    size_t total = (size_t)rows * row_bytes;
+    hipLaunchKernelGGL(k_copy_strided_b8, dim3(copy_grid(total / 16 + 1)),
+                       dim3(256), 0, stream, (const uint8_t*)src, src_stride,
+                       (uint8_t*)dst, dst_stride, rows, row_bytes);
   }
   return hipGetLastError();
 }

@@ -33,19 +33,40 @@ static BufferRef parse_buffer(py::handle obj, bool writable) {
     size_t itemsize = std::stoul(typestr.substr(2));
     size_t n = 1;
     for (auto d : shape) n *= d.cast<size_t>();
+    uint64_t rows = 0, row_bytes = 0, row_stride = 0;
     if (cai.contains("strides") && !cai["strides"].is_none()) {
-      // Require C-contiguity: strides must match row-major layout.
       py::tuple strides = cai["strides"].cast<py::tuple>();
       size_t expect = itemsize;
+      bool contiguous = true;
       for (ssize_t i = (ssize_t)shape.size() - 1; i >= 0; --i) {
-        if (strides[i].cast<size_t>() != expect)
-          throw std::invalid_argument(
-              "device buffer must be C-contiguous for zero-copy messaging");
+        if (strides[i].cast<size_t>() != expect) contiguous = false;
         expect *= shape[i].cast<size_t>();
+      }
+      if (!contiguous) {
+        // 2D row-strided layout is supported natively (pack/unpack in the
+        // pull kernel): last dim dense, leading dim strided.
+        if (shape.size() == 2 &&
+            strides[1].cast<size_t>() == itemsize) {
+          rows = shape[0].cast<uint64_t>();
+          row_bytes = shape[1].cast<uint64_t>() * itemsize;
+          row_stride = strides[0].cast<uint64_t>();
+          if (row_stride < row_bytes)
+            throw std::invalid_argument(
+                "overlapping strided device buffer is not supported");
+        } else {
+          throw std::invalid_argument(
+              "device buffer must be contiguous or 2D row-strided "
+              "(last dim dense) for zero-copy messaging");
+        }
       }
     }
     ref.ptr = (uint8_t*)ptr;
     ref.size = n * itemsize;
+    if (rows > 1) {
+      ref.rows = rows;
+      ref.row_bytes = row_bytes;
+      ref.stride = row_stride;
+    }
     ref.device = gpu::device_of((const void*)ptr);
     if (ref.device < 0)
       throw std::invalid_argument(

@@ -86,20 +86,33 @@ class _CudaShim:
     __slots__ = ("_t", "__cuda_array_interface__")
 
     def __init__(self, t) -> None:
-        if not t.is_contiguous():
+        es = t.element_size()
+        if t.is_contiguous():
+            nbytes = t.numel() * es
+            self.__cuda_array_interface__ = {
+                "data": (t.data_ptr(), False),
+                "shape": (nbytes,),
+                "typestr": "|u1",
+                "strides": None,
+                "version": 2,
+            }
+        elif t.dim() == 2 and t.stride(1) == 1:
+            # 2D row-strided slice: moved natively by the strided
+            # pack/unpack kernels — no .contiguous() staging copy.
+            r, c = t.shape
+            self.__cuda_array_interface__ = {
+                "data": (t.data_ptr(), False),
+                "shape": (r, c * es),
+                "typestr": "|u1",
+                "strides": (t.stride(0) * es, 1),
+                "version": 2,
+            }
+        else:
             raise ValueError(
-                "device message buffers must be contiguous for zero-copy "
-                "transfer (call .contiguous() first)"
+                "device message buffers must be contiguous or 2D row-strided "
+                "(call .contiguous() first)"
             )
         self._t = t
-        nbytes = t.numel() * t.element_size()
-        self.__cuda_array_interface__ = {
-            "data": (t.data_ptr(), False),
-            "shape": (nbytes,),
-            "typestr": "|u1",
-            "strides": None,
-            "version": 2,
-        }
 
 
 def _norm_buffer(buf: Any) -> Any:
