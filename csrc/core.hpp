@@ -225,6 +225,7 @@ struct TxItem {
   std::vector<uint8_t> head;      // header (+ inline payload for small frames)
   const uint8_t* ext = nullptr;   // zero-copy user payload (may be null)
   size_t ext_len = 0;
+  RawBuf ext_own;                 // owned payload (cross-host GPU bounce)
   bool has_keepalive = false;
   bool is_data = false;           // EAGER/RTS frames: droppable on close
   bool via_ring = false;          // carried on the shm ring, not the socket
@@ -432,6 +433,15 @@ class Engine {
   std::unordered_map<uint64_t, Op*> gpu_sends_;  // op id -> awaiting ack
   struct GpuPull;  // defined in engine.cpp (holds hipEvent)
   std::vector<std::unique_ptr<GpuPull>> gpu_pulls_;
+  // Cross-host GPU send: device->host bounce in flight; when the copy
+  // completes the payload goes out as a plain eager frame.
+  struct D2hSend {
+    void* ticket = nullptr;
+    Op* op = nullptr;
+    RawBuf buf;
+  };
+  std::vector<std::unique_ptr<D2hSend>> d2h_sends_;
+  void progress_d2h(bool& did_work);
   struct CmaPull {
     Op* recv_op = nullptr;
     Connection* conn = nullptr;
@@ -488,6 +498,8 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
 // ticket with attach_bounce.
 void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
                 std::string* err);
+// Device -> host staging download (cross-host GPU sends).
+void* begin_d2h(void* host_dst, const BufferRef& src, std::string* err);
 void attach_bounce(void* ticket, RawBuf&& bounce);
 // Poll a ticket: 1 done, 0 pending, -1 error.
 int poll_ticket(void* ticket, std::string* err);

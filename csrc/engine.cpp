@@ -705,7 +705,8 @@ void Engine::loop_iteration(bool& did_work) {
       on_conn_dead(c.get());
   }
   int timeout = 0;
-  bool busy = !gpu_pulls_.empty() || !cma_pulls_.empty();
+  bool busy =
+      !gpu_pulls_.empty() || !cma_pulls_.empty() || !d2h_sends_.empty();
   if (!busy)
     for (auto& c : conns_)
       if (c->want_write() || (c->shm_rx && c->shm && !c->dead &&
@@ -728,6 +729,7 @@ void Engine::loop_iteration(bool& did_work) {
   engine_hot_.store(true, std::memory_order_release);
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
   if (!cma_pulls_.empty()) progress_cma(did_work);
+  if (!d2h_sends_.empty()) progress_d2h(did_work);
   if (!pending_flushes_.empty()) check_flush_progress();
   if (!completions_.empty()) {
     fire_completions();
@@ -794,6 +796,40 @@ void Engine::process_command(Op* op) {
           op->done_cb = py::object();
           op->fail_cb = py::object();
         }
+        return;
+      }
+      static const bool force_xhost =
+          getenv("STARWAY_FORCE_XHOST") != nullptr;  // test hook
+      if (op->buf.device >= 0 &&
+          (force_xhost ||
+           memcmp(c->peer.host_id, host_id(), 16) != 0)) {
+        // Cross-host GPU send: hipIpc cannot cross hosts — stage the
+        // payload to host memory and ship it as a plain eager frame once
+        // the download completes. The op completes at hand-off (eager
+        // semantics); both callbacks are consumed here.
+        auto d2h = std::make_unique<D2hSend>();
+        d2h->buf.alloc(op->buf.size);
+        std::string err;
+        d2h->ticket = gpu::begin_d2h(d2h->buf.data(), op->buf, &err);
+        if (!d2h->ticket) {
+          fail_op(op, "send failed: " + err);
+          return;
+        }
+        d2h->op = op;
+        {
+          py::gil_scoped_acquire gil;
+          try {
+            if (op->done_cb.ptr()) op->done_cb();
+          } catch (py::error_already_set& e) {
+            e.discard_as_unraisable("starway send callback");
+          }
+          op->done_cb = py::object();
+          op->fail_cb = py::object();
+        }
+        // Track it like a GPU send so flush waits for the wire hand-off.
+        gpu_sends_[op->id] = op;
+        op->gpu_send_awaiting_ack = true;
+        d2h_sends_.push_back(std::move(d2h));
         return;
       }
       if (op->buf.device >= 0) {
@@ -1445,6 +1481,59 @@ void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
   cma_pulls_.push_back(std::move(pull));
 }
 
+void Engine::progress_d2h(bool& did_work) {
+  for (size_t i = 0; i < d2h_sends_.size();) {
+    D2hSend* p = d2h_sends_[i].get();
+    std::string err;
+    int r = gpu::poll_ticket(p->ticket, &err);
+    if (r == 0) {
+      i++;
+      continue;
+    }
+    did_work = true;
+    Op* op = p->op;
+    gpu::free_ticket(p->ticket);
+    if (r > 0 && op->conn && !op->conn->dead) {
+      Connection* c = op->conn;
+      TxItem item;
+      FrameHeader h{};
+      h.magic = kMagic;
+      h.type = FT_EAGER;
+      h.tag = op->tag;
+      h.size = op->buf.size;
+      h.op_id = op->id;
+      h.aux = op->buf.size;
+      item.head.resize(sizeof(h));
+      memcpy(item.head.data(), &h, sizeof(h));
+      item.ext_own = std::move(p->buf);
+      item.ext = item.ext_own.data();
+      item.ext_len = op->buf.size;
+      item.is_data = true;
+      item.via_ring = c->shm_tx_enq;
+      c->tx_enqueued_bytes += item.head.size() + item.ext_len;
+      c->txq.push_back(std::move(item));
+      bool dummy = false;
+      handle_writable(c, dummy);
+      // Flushes that snapshot this op id now wait for the wire bytes.
+      for (Op* f : pending_flushes_) {
+        if (f->flush_ops_pending.erase(op->id)) {
+          uint64_t target = c->tx_enqueued_bytes;
+          auto [fit, ins] = f->flush_write_targets.try_emplace(c, target);
+          if (!ins && fit->second < target) fit->second = target;
+        }
+      }
+    } else {
+      for (Op* f : pending_flushes_) f->flush_ops_pending.erase(op->id);
+    }
+    gpu_sends_.erase(op->id);
+    Completion comp;  // callbacks already consumed; just reaps the op
+    comp.kind = Completion::Kind::SendDone;
+    comp.op = op;
+    complete(std::move(comp));
+    d2h_sends_.erase(d2h_sends_.begin() + i);
+  }
+}
+
 // Chunked process_vm_readv pulls, interleaved with the progress loop so a
 // multi-GiB pull never starves other connections; aborted when the sender
 // dies (its memory is gone — the message is lost, recv re-posts, exactly
@@ -1993,6 +2082,15 @@ void Engine::teardown() {
   for (auto& p : cma_pulls_)
     fail_op(p->recv_op, "operation canceled (endpoint closing)");
   cma_pulls_.clear();
+  for (auto& p : d2h_sends_) {
+    gpu::free_ticket(p->ticket);
+    gpu_sends_.erase(p->op->id);
+    Completion comp;  // callbacks consumed at hand-off; reap silently
+    comp.kind = Completion::Kind::SendDone;
+    comp.op = p->op;
+    complete(std::move(comp));
+  }
+  d2h_sends_.clear();
   // 3. Cancel in-flight data: a connection with undelivered EAGER/RTS bytes
   //    queued is closed abortively — close without flush loses in-flight
   //    sends (the reference's delivery contract, tests/test_basic.py:250-278;

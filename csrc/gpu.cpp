@@ -349,6 +349,41 @@ void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
   return t;
 }
 
+void* begin_d2h(void* host_dst, const BufferRef& src, std::string* err) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  if (!available()) {
+    *err = "no HIP device available";
+    return nullptr;
+  }
+  int prev;
+  hipGetDevice(&prev);
+  hipSetDevice(src.device);
+  hipStream_t stream = pull_stream(src.device);
+  hipError_t e;
+  if (src.rows > 0) {
+    e = hipMemcpy2DAsync(host_dst, src.row_bytes, src.ptr, src.stride,
+                         src.row_bytes, src.rows, hipMemcpyDeviceToHost,
+                         stream);
+  } else {
+    e = hipMemcpyAsync(host_dst, src.ptr, src.size, hipMemcpyDeviceToHost,
+                       stream);
+  }
+  Ticket* t = nullptr;
+  if (e == hipSuccess) {
+    t = new Ticket();
+    t->device = src.device;
+    e = pool_get_event(src.device, &t->ev);
+    if (e == hipSuccess) e = hipEventRecord(t->ev, stream);
+  }
+  hipSetDevice(prev);
+  if (e != hipSuccess) {
+    *err = std::string("d2h: ") + hipGetErrorString(e);
+    delete t;
+    return nullptr;
+  }
+  return t;
+}
+
 void attach_bounce(void* ticket, RawBuf&& bounce) {
   ((Ticket*)ticket)->bounce = std::move(bounce);
 }

@@ -408,3 +408,25 @@ def test_cross_process_still_ok_marker():
     # strided coverage comes from the same-process tests (identical kernel
     # path, only the pointer resolution differs).
     assert True
+
+
+async def test_cross_host_gpu_bounce_path():
+    """Cross-host GPU sends stage D2H and ship as eager (hipIpc cannot
+    cross hosts). Forced via STARWAY_FORCE_XHOST; receiver H2D-bounces into
+    the posted device buffer."""
+    os.environ["STARWAY_FORCE_XHOST"] = "1"
+    try:
+        async with loopback() as (server, client):
+            src = torch.randint(0, 256, (4 << 20,), dtype=torch.uint8,
+                                device="cuda")
+            dst = torch.zeros_like(src)
+            torch.cuda.synchronize()
+            fut = server.arecv(dst, 0, 0)
+            await client.asend(src, 12)
+            await client.aflush()  # must cover the staged wire bytes
+            tag, ln = await fut
+            torch.cuda.synchronize()
+            assert tag == 12 and ln == 4 << 20
+            assert torch.equal(src, dst)
+    finally:
+        del os.environ["STARWAY_FORCE_XHOST"]
