@@ -53,3 +53,17 @@ def pytest_pyfunc_call(pyfuncitem):
         asyncio.run(asyncio.wait_for(fn(**kwargs), timeout=ASYNC_TEST_TIMEOUT))
         return True
     return None
+
+
+@pytest.fixture
+def port():
+    """A free TCP port from the kernel's ephemeral range (random fixed-range
+    picks collided intermittently across the suite). Our listener sets
+    SO_REUSEADDR, so the close->rebind window is safe."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p

@@ -22,10 +22,6 @@ SERVER_ADDR = "127.0.0.1"
 INFLIGHT_BYTES = 1024 * 1024 * 1024  # large enough to be in flight
 
 
-@pytest.fixture
-def port():
-    return random.randint(10000, 50000)
-
 
 @contextlib.asynccontextmanager
 async def gen_server_client(port):

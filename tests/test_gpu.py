@@ -24,10 +24,6 @@ import starway_amd as sw  # noqa: E402
 from starway_amd import _core  # noqa: E402
 
 
-@pytest.fixture
-def port():
-    return random.randint(10000, 50000)
-
 
 @contextlib.asynccontextmanager
 async def loopback():

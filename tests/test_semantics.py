@@ -16,10 +16,6 @@ SERVER_ADDR = "127.0.0.1"
 FULL = (1 << 64) - 1
 
 
-@pytest.fixture
-def port():
-    return random.randint(10000, 50000)
-
 
 @contextlib.asynccontextmanager
 async def pair(port):

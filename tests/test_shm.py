@@ -13,10 +13,6 @@ from starway_amd import Client, Server
 SERVER_ADDR = "127.0.0.1"
 
 
-@pytest.fixture
-def port():
-    return random.randint(10000, 50000)
-
 
 async def test_shm_activates_same_host(port):
     server = Server()

@@ -16,10 +16,6 @@ from starway_amd import Client, Server
 SERVER_ADDR = "127.0.0.1"
 
 
-@pytest.fixture
-def port():
-    return random.randint(10000, 50000)
-
 
 def test_multithreaded_send_recv(port):
     """4 sender threads (raw callback API) x 500 msgs against one server;
