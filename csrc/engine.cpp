@@ -1316,8 +1316,11 @@ void Engine::begin_eager(Connection* c) {
     Op* r = *it;
     if ((tag & r->tag_mask) == (r->tag & r->tag_mask)) {
       posted_recvs_.erase(it);
-      if (msg_len > r->buf.size) {
-        // Truncation: consume + fail (UCX MESSAGE_TRUNCATED analog).
+      if (msg_len > r->buf.size ||
+          (r->buf.rows > 0 && msg_len != r->buf.size)) {
+        // Truncation / strided-geometry mismatch: consume + fail
+        // (UCX MESSAGE_TRUNCATED analog; a strided window needs the exact
+        // message size or partial rows would smear).
         c->rx_truncated = true;
         c->rx_recv_op = r;
       } else {
