@@ -215,6 +215,10 @@ struct EndpointInfo {
   std::vector<std::pair<std::string, std::string>> transports;
   Connection* conn = nullptr;  // nulled when the connection dies
   class Engine* owner = nullptr;
+  // Peer traits snapshotted at accept time (safe to read from Python
+  // threads; Connection fields are engine-thread-only).
+  bool peer_has_gpu = false;
+  bool peer_same_proc = false;
 };
 
 // ---------------------------------------------------------------------------
@@ -401,7 +405,10 @@ class Engine {
   void wake();
   Connection* make_listener(const std::string& addr, int port);
 
-  double perf_model(Connection* c, uint64_t msg_size) const;
+  double perf_model(bool peer_gpu, bool same_proc, uint64_t msg_size) const;
+  // Client-side peer traits (bit0 has_gpu, bit1 same_proc; -1 unknown),
+  // published at HELLO so evaluate_perf never touches engine-owned state.
+  std::atomic<int> client_peer_traits_{-1};
 
  public:
   Mode mode_;

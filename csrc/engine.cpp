@@ -489,26 +489,23 @@ std::vector<std::shared_ptr<EndpointInfo>> Engine::list_clients() {
   return eps_;
 }
 
-double Engine::perf_model(Connection* c, uint64_t msg_size) const {
+double Engine::perf_model(bool peer_gpu, bool same_proc,
+                          uint64_t msg_size) const {
   // Analytic transfer-time model, the ucp_ep_evaluate_perf analog
-  // (reference main.cpp:452-467). Constants from xGMI topology (7 links x
-  // ~153 GB/s per MI355X) and localhost TCP measurements; recalibrated once
-  // rocprof evidence lands (profiles/).
-  // Calibrated against round-1 MI355X measurements (profiles/
-  // r01_sweep_summary.md): device path ~33 us base latency; same-GPU
-  // delivery ~2 TB/s; cross-GPU is xGMI-link-bound (~140 GB/s sustained
-  // per pair). CPU path: localhost TCP ~80 us half-RTT, ~3 GB/s.
-  bool peer_gpu = c && c->peer.has_gpu;
+  // (reference main.cpp:452-467), calibrated against round-1 MI355X
+  // measurements (profiles/r01_sweep_summary.md): device path ~33 us base
+  // latency; same-GPU delivery ~2 TB/s; cross-GPU is xGMI-link-bound
+  // (~140 GB/s sustained per pair); CPU path ~80 us / ~3 GB/s (tcp) with
+  // shm/CMA improving both on the same host.
   bool self_gpu = gpu::available();
   double lat, bw;
   if (self_gpu && peer_gpu) {
-    bool same_proc = c && memcmp(c->peer.uuid, process_uuid(), 16) == 0;
-    lat = 33e-6;  // RTS over TCP + pull-kernel launch + event poll
+    lat = 33e-6;  // RTS + pull-kernel launch + event poll
     // Same process usually means same device (loopback benches); across
     // processes assume the conservative single-link xGMI figure.
     bw = same_proc ? 2.0e12 : 140e9;
   } else {
-    lat = 80e-6;  // localhost TCP eager
+    lat = 80e-6;
     bw = 3e9;
   }
   return lat + (double)msg_size / bw;
@@ -518,8 +515,11 @@ double Engine::evaluate_perf(std::shared_ptr<EndpointInfo> ep,
                              uint64_t msg_size) {
   if (status_.load(std::memory_order_acquire) != 2)
     throw std::runtime_error("evaluate_perf: endpoint not running");
-  Connection* c = ep ? ep->conn : (conns_.empty() ? nullptr : conns_[0].get());
-  return perf_model(c, msg_size);
+  if (ep) return perf_model(ep->peer_has_gpu, ep->peer_same_proc, msg_size);
+  int traits = client_peer_traits_.load(std::memory_order_acquire);
+  bool peer_gpu = traits > 0 && (traits & 1);
+  bool same_proc = traits > 0 && (traits & 2);
+  return perf_model(peer_gpu, same_proc, msg_size);
 }
 
 // ---- engine thread --------------------------------------------------------
@@ -1148,6 +1148,10 @@ void Engine::on_frame_payload(Connection* c) {
       }
       c->peer = pi;
       c->hello_received = true;
+      client_peer_traits_.store(
+          (pi.has_gpu ? 1 : 0) |
+              (memcmp(pi.uuid, process_uuid(), 16) == 0 ? 2 : 0),
+          std::memory_order_release);
       on_hello(c);
       break;
     }
@@ -1251,6 +1255,8 @@ void Engine::on_hello(Connection* c) {
     ep->remote_port = c->remote_port;
     ep->conn = c;
     ep->owner = this;
+    ep->peer_has_gpu = c->peer.has_gpu;
+    ep->peer_same_proc = memcmp(c->peer.uuid, process_uuid(), 16) == 0;
     ep->transports.emplace_back("tcp", "sock");
     if (gpu::available() && c->peer.has_gpu) {
       bool same_proc = memcmp(c->peer.uuid, process_uuid(), 16) == 0;
