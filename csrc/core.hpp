@@ -195,6 +195,9 @@ struct Op {
   uint64_t recv_len = 0;          // actual message length for recv completion
   uint64_t recv_sender_tag = 0;
   bool gpu_send_awaiting_ack = false;
+  // Op lifetime is owned by the d2h staging list (progress_d2h reaps it);
+  // connection-death cleanup must not delete it a second time.
+  bool owned_by_d2h = false;
   // Flush bookkeeping (UCX worker/ep-flush semantics: complete when all
   // bytes queued before the flush are written to the wire AND all GPU
   // rendezvous sends posted before it are delivered):

@@ -829,6 +829,7 @@ void Engine::process_command(Op* op) {
         // Track it like a GPU send so flush waits for the wire hand-off.
         gpu_sends_[op->id] = op;
         op->gpu_send_awaiting_ack = true;
+        op->owned_by_d2h = true;  // progress_d2h reaps; see on_conn_dead
         d2h_sends_.push_back(std::move(d2h));
         return;
       }
@@ -1996,10 +1997,13 @@ void Engine::on_conn_dead(Connection* c) {
       i++;
     }
   }
-  // GPU sends routed to this conn will never be acked.
+  // GPU sends routed to this conn will never be acked. Ops owned by the
+  // d2h staging list are only unregistered here — progress_d2h reaps them
+  // (double-delete hazard otherwise).
   for (auto it = gpu_sends_.begin(); it != gpu_sends_.end();) {
     if (it->second->conn == c) {
-      fail_op(it->second, "send failed: connection reset");
+      if (!it->second->owned_by_d2h)
+        fail_op(it->second, "send failed: connection reset");
       it = gpu_sends_.erase(it);
     } else {
       ++it;
