@@ -2095,6 +2095,21 @@ void Engine::teardown() {
   for (auto& p : cma_pulls_)
     fail_op(p->recv_op, "operation canceled (endpoint closing)");
   cma_pulls_.clear();
+  // The async D2H copies write into the staging RawBufs below — wait for
+  // their tickets (bounded) before freeing them.
+  {
+    auto d2h_deadline =
+        std::chrono::steady_clock::now() + std::chrono::seconds(10);
+    bool pending = true;
+    while (pending && std::chrono::steady_clock::now() < d2h_deadline) {
+      pending = false;
+      for (auto& p : d2h_sends_) {
+        std::string err;
+        if (gpu::poll_ticket(p->ticket, &err) == 0) pending = true;
+      }
+      if (pending) sched_yield();
+    }
+  }
   for (auto& p : d2h_sends_) {
     gpu::free_ticket(p->ticket);
     gpu_sends_.erase(p->op->id);
