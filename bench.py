@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--msg-bytes", type=int, default=256 * 1024 * 1024,
                    help="per-rank outbound bytes per step (split across peers)")
+    p.add_argument("--chunks", type=int, default=4,
+                   help="tagged messages per peer per step; >1 pipelines "
+                        "per-message control latency behind the copy kernels")
     p.add_argument("--lat-iters", type=int, default=200,
                    help="64B pingpong iterations for the latency probe")
     p.add_argument("--device", default=None,
@@ -54,8 +57,9 @@ def parse_args():
     return p.parse_args()
 
 
-def make_tag(src_rank: int, step: int) -> int:
-    return (1 << 60) | (src_rank << 32) | (step & 0xFFFFFFFF)
+def make_tag(src_rank: int, step: int, chunk: int = 0) -> int:
+    return (1 << 60) | (src_rank << 40) | ((chunk & 0xFF) << 32) | (
+        step & 0xFFFFFFFF)
 
 
 LAT_TAG_BASE = 1 << 59
@@ -109,8 +113,14 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
         clients = {0: c}
         peers = [0]
 
+    chunks = max(1, min(args.chunks, per_peer))
+    bounds = [per_peer * k // chunks for k in range(chunks + 1)]
     send_bufs = {j: alloc(per_peer, fill=(rank * 31 + j) % 251) for j in peers}
     recv_bufs = {j: alloc(per_peer) for j in peers}
+    send_views = {j: [send_bufs[j][bounds[k]:bounds[k + 1]]
+                      for k in range(chunks)] for j in peers}
+    recv_views = {j: [recv_bufs[j][bounds[k]:bounds[k + 1]]
+                      for k in range(chunks)] for j in peers}
     sync_device()
 
     mesh = None
@@ -141,12 +151,12 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
             mesh.synchronize()
             return
         recvs = [
-            server.arecv(recv_bufs[j], make_tag(j, step_idx), full_mask)
-            for j in peers
+            server.arecv(recv_views[j][k], make_tag(j, step_idx, k), full_mask)
+            for j in peers for k in range(chunks)
         ]
         sends = [
-            clients[j].asend(send_bufs[j], make_tag(rank, step_idx))
-            for j in peers
+            clients[j].asend(send_views[j][k], make_tag(rank, step_idx, k))
+            for j in peers for k in range(chunks)
         ]
         await asyncio.gather(*sends, *recvs)
 
@@ -242,7 +252,10 @@ def main() -> int:
     elapsed, half_rtt_us = asyncio.run(run_rank(args, rank, world, device, dist))
 
     if rank == 0:
-        total_bytes = world * args.msg_bytes * args.steps
+        n_peers = max(1, world - 1)
+        per_peer = args.msg_bytes // n_peers if world > 1 else args.msg_bytes
+        moved_per_rank = per_peer * (n_peers if world > 1 else 1)
+        total_bytes = world * moved_per_rank * args.steps
         gbps = total_bytes / elapsed / 1e9
         result = {
             "metric": "tagged_allpairs_bandwidth",
@@ -261,6 +274,7 @@ def main() -> int:
                 "model": "tagged-pingpong-allpairs",
                 "message_bytes_per_rank": args.msg_bytes,
                 "message_bytes_per_peer": args.msg_bytes // max(1, world - 1),
+                "chunks_per_peer": args.chunks,
                 "endpoints": world,
                 "device": device,
                 "transport": args.transport,
