@@ -152,9 +152,16 @@ struct BufferRef {
 struct RawBuf {
   std::unique_ptr<uint8_t[]> p;
   size_t n = 0;
-  void alloc(size_t s) {
-    p.reset(new uint8_t[s]);  // default-init: uninitialized for uint8_t
+  // Returns false on allocation failure (a peer announcing a huge message
+  // must not std::terminate the engine thread).
+  bool alloc(size_t s) {
+    p.reset(new (std::nothrow) uint8_t[s]);
+    if (!p && s > 0) {
+      n = 0;
+      return false;
+    }
     n = s;
+    return true;
   }
   uint8_t* data() { return p.get(); }
   size_t size() const { return n; }
@@ -281,6 +288,7 @@ struct Connection {
   std::vector<uint8_t> rx_discard;   // truncation sink
   uint64_t rx_msg_remaining = 0;
   bool rx_truncated = false;
+  bool rx_discarding = false;  // consume current message into the sink
   // GPU eager-recv host bounce (posted recv buffer is on device):
   RawBuf rx_gpu_bounce;
 

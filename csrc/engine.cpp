@@ -538,7 +538,16 @@ void Engine::thread_main() {
   bool did_work = false;
   while (status_.load(std::memory_order_acquire) == 2) {
     did_work = false;
-    loop_iteration(did_work);
+    try {
+      loop_iteration(did_work);
+    } catch (const std::exception& e) {
+      // Never let an engine-thread exception reach std::terminate: report
+      // and shut the endpoint down (pending ops get canceled in teardown).
+      fprintf(stderr, "[starway] engine error: %s — closing endpoint\n",
+              e.what());
+      status_.store(3, std::memory_order_release);
+      break;
+    }
     if (did_work) {
       idle_iters_ = 0;
     } else {
@@ -808,7 +817,10 @@ void Engine::process_command(Op* op) {
         // the download completes. The op completes at hand-off (eager
         // semantics); both callbacks are consumed here.
         auto d2h = std::make_unique<D2hSend>();
-        d2h->buf.alloc(op->buf.size);
+        if (!d2h->buf.alloc(op->buf.size)) {
+          fail_op(op, "send failed: staging allocation failed");
+          return;
+        }
         std::string err;
         d2h->ticket = gpu::begin_d2h(d2h->buf.data(), op->buf, &err);
         if (!d2h->ticket) {
@@ -1041,7 +1053,7 @@ void Engine::handle_stream(Connection* c, bool from_ring, bool& did_work) {
           } else {
             dst = c->rx_recv_op->buf.ptr + done;
           }
-        } else if (c->rx_unexp) {
+        } else if (c->rx_unexp && !c->rx_discarding) {
           dst = (c->rx_unexp->redirect_dst ? c->rx_unexp->redirect_dst
                                            : c->rx_unexp->data.data()) +
                 done;
@@ -1316,6 +1328,7 @@ void Engine::begin_eager(Connection* c) {
   c->rx_recv_op = nullptr;
   c->rx_unexp = nullptr;
   c->rx_truncated = false;
+  c->rx_discarding = false;
   c->rx_msg_remaining = msg_len;
 
   // Match against posted recvs in FIFO order.
@@ -1332,7 +1345,17 @@ void Engine::begin_eager(Connection* c) {
         c->rx_recv_op = r;
       } else {
         c->rx_recv_op = r;
-        if (r->buf.device >= 0) c->rx_gpu_bounce.alloc(msg_len);
+        if (r->buf.device >= 0 && !c->rx_gpu_bounce.alloc(msg_len)) {
+          // Cannot stage the host bounce: fail the recv, then consume and
+          // discard the stream bytes (truncation machinery reused with the
+          // op already failed).
+          c->rx_recv_op = nullptr;
+          fail_op(r, "receive failed: host bounce allocation failed");
+          c->rx_state =
+              msg_len ? Connection::RxState::Payload : Connection::RxState::Header;
+          c->rx_discarding = true;
+          return;
+        }
       }
       c->rx_recv_op->recv_sender_tag = tag;
       c->rx_recv_op->recv_len = msg_len;
@@ -1345,7 +1368,13 @@ void Engine::begin_eager(Connection* c) {
     um->tag = tag;
     um->size = msg_len;
     um->conn = c;
-    um->data.alloc(msg_len);
+    if (!um->data.alloc(msg_len)) {
+      // Cannot stage it: drop the connection rather than the process.
+      SW_DBG("unexpected staging alloc failed (%llu bytes)",
+             (unsigned long long)msg_len);
+      on_conn_dead(c);
+      return;
+    }
     c->rx_unexp = um.get();
     unexpected_.push_back(std::move(um));
   }
@@ -1359,6 +1388,10 @@ void Engine::begin_eager(Connection* c) {
 void Engine::finish_eager_into_recv(Connection* c) {
   c->rx_state = Connection::RxState::Header;
   c->rx_got = 0;
+  if (c->rx_discarding) {
+    c->rx_discarding = false;
+    return;
+  }
   if (c->rx_recv_op) {
     Op* r = c->rx_recv_op;
     c->rx_recv_op = nullptr;
