@@ -230,6 +230,83 @@ def run_raw_latency(device: str) -> list[dict]:
     return results
 
 
+def run_raw_rate(device: str, msg_bytes: int = 1024, depth: int = 64,
+                 total: int = 20000) -> list[dict]:
+    """Raw-callback small-message rate (no asyncio): `depth` messages kept
+    in flight; measures sustained msgs/s to locate the engine-side cap."""
+    import threading
+
+    import starway_amd as sw
+
+    server, client = sw.Server(), sw.Client()
+    addr = server.listen_address()
+    ev = threading.Event()
+    client._client.connect_address(addr, lambda s: ev.set())
+    assert ev.wait(10)
+
+    sbufs = [_alloc(msg_bytes, device, fill=i % 251) for i in range(depth)]
+    rbufs = [_alloc(msg_bytes, device) for i in range(depth)]
+    done = threading.Semaphore(0)
+    remaining = [total]   # completions outstanding
+    posted = [0]          # recvs posted so far
+    lock = threading.Lock()
+
+    def recv_cb(slot):
+        def cb(tag, length):
+            repost = False
+            with lock:
+                remaining[0] -= 1
+                if remaining[0] <= 0:
+                    done.release()
+                    return
+                if posted[0] < total:
+                    posted[0] += 1
+                    repost = True
+            if repost:
+                server._server.recv(rbufs[slot], 0, 0, recv_cb(slot),
+                                    lambda e: done.release())
+        return cb
+
+    t0 = time.perf_counter()
+    for i in range(depth):
+        with lock:
+            posted[0] += 1
+        server._server.recv(rbufs[i], 0, 0, recv_cb(i),
+                            lambda e: done.release())
+
+    sent = [0]
+
+    def send_more(slot):
+        def cb():
+            with lock:
+                if sent[0] >= total:
+                    return
+                sent[0] += 1
+            client._client.send(sbufs[slot], sent[0], cb,
+                                lambda e: None)
+        return cb
+
+    for i in range(depth):
+        with lock:
+            sent[0] += 1
+        client._client.send(sbufs[i], i, send_more(i), lambda e: None)
+
+    assert done.acquire(timeout=120)
+    dt = time.perf_counter() - t0
+    rate = total / dt
+    print(f"raw rate: {rate:,.0f} msgs/s ({msg_bytes} B x {total}, depth "
+          f"{depth}, {dt:.2f}s)", flush=True)
+
+    ev2 = threading.Event()
+    client._client.close(lambda: ev2.set())
+    ev2.wait(10)
+    ev3 = threading.Event()
+    server._server.close(lambda: ev3.set())
+    ev3.wait(10)
+    return [{"mode": "raw-rate", "msgs_per_s": rate, "msg_bytes": msg_bytes,
+             "depth": depth}]
+
+
 def main():
     if os.environ.get("SW_FH"):
         import faulthandler
@@ -240,6 +317,8 @@ def main():
     ap.add_argument("--cross-process", action="store_true")
     ap.add_argument("--raw", action="store_true",
                     help="raw-callback latency floor (no asyncio)")
+    ap.add_argument("--rate", action="store_true",
+                    help="raw-callback small-message rate probe")
     ap.add_argument("--out", default="gpurun_out/sweep.json")
     args = ap.parse_args()
     if args.device is None:
@@ -250,7 +329,9 @@ def main():
         except ImportError:
             args.device = "cpu"
 
-    if args.raw:
+    if args.rate:
+        results = run_raw_rate(args.device)
+    elif args.raw:
         results = run_raw_latency(args.device)
     else:
         results = asyncio.run(run_sweep(args))
