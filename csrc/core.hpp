@@ -451,6 +451,18 @@ class Engine {
   std::unordered_map<uint64_t, Op*> gpu_sends_;  // op id -> awaiting ack
   struct GpuPull;  // defined in engine.cpp (holds hipEvent)
   std::vector<std::unique_ptr<GpuPull>> gpu_pulls_;
+  // Small device pulls collected during a drain, launched batched at the
+  // end of the loop iteration (one kernel + one event for up to 8).
+  struct SmallPull {
+    RtsDesc rts;
+    Op* recv_op;
+    Connection* conn;
+    uint64_t sender_op_id;
+    uint64_t tag;
+    uint64_t size;
+  };
+  std::vector<SmallPull> pending_small_pulls_;
+  void flush_small_pulls();
   // Cross-host GPU send: device->host bounce in flight; when the copy
   // completes the payload goes out as a plain eager frame.
   struct D2hSend {
@@ -511,6 +523,16 @@ bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err);
 // device). Returns an opaque ticket (hipEvent) or null on error.
 void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
                  std::string* err);
+// Batched small-message pull: one launch + one event for up to 8 messages
+// targeting the same destination device. Each entry must be a contiguous
+// device->device transfer. Returns one shared ticket.
+struct PullReq {
+  RtsDesc rts;
+  uint8_t* dst_ptr;
+  int dst_device;
+  uint64_t size;
+};
+void* begin_pull_multi(const PullReq* reqs, int n, std::string* err);
 // Begin an async host->device upload into a device buffer. The host memory
 // must stay valid until the ticket completes; hand its ownership to the
 // ticket with attach_bounce.

@@ -252,6 +252,9 @@ struct Engine::GpuPull {
   uint64_t sender_op_id = 0;      // 0 => no RECV_DONE ack needed (h2d bounce)
   uint64_t tag = 0;
   uint64_t len = 0;
+  // Batched small-message pull: completions for every message sharing the
+  // one ticket (recv_op is null in that case).
+  std::vector<Engine::SmallPull> batch;
 };
 
 // ---------------------------------------------------------------------------
@@ -736,6 +739,10 @@ void Engine::loop_iteration(bool& did_work) {
   }
   poll_sockets(timeout, did_work);
   engine_hot_.store(true, std::memory_order_release);
+  if (!pending_small_pulls_.empty()) {
+    flush_small_pulls();
+    did_work = true;
+  }
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
   if (!cma_pulls_.empty()) progress_cma(did_work);
   if (!d2h_sends_.empty()) progress_d2h(did_work);
@@ -1485,6 +1492,16 @@ void Engine::start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
     fail_op(recv_op, "receive failed: " + err);
     return;
   }
+  // Small contiguous device->device messages are batched: collected here,
+  // launched as one multi-copy kernel + one event at the end of the loop
+  // iteration (the per-launch cost dominates small-message rate).
+  static const uint64_t kMultiMax = env_u64("STARWAY_MULTI_COPY_MAX", 65536);
+  if (size > 0 && size <= kMultiMax && recv_op->buf.device >= 0 &&
+      recv_op->buf.rows == 0 && rts.src_rows == 0) {
+    pending_small_pulls_.push_back(
+        SmallPull{rts, recv_op, c, sender_op, tag, size});
+    return;
+  }
   std::string err;
   void* ticket = gpu::begin_pull(rts, recv_op->buf, size, &err);
   if (!ticket) {
@@ -1501,6 +1518,58 @@ void Engine::start_gpu_pull(Op* recv_op, const RtsDesc& rts, uint64_t tag,
   pull->tag = tag;
   pull->len = size;
   gpu_pulls_.push_back(std::move(pull));
+}
+
+void Engine::flush_small_pulls() {
+  // Group by destination device; launch batches of up to 8 per kernel. A
+  // lone pull goes through the ordinary single path.
+  while (!pending_small_pulls_.empty()) {
+    int dev = pending_small_pulls_[0].recv_op->buf.device;
+    gpu::PullReq reqs[8];
+    SmallPull items[8];
+    int n = 0;
+    for (size_t i = 0; i < pending_small_pulls_.size() && n < 8;) {
+      if (pending_small_pulls_[i].recv_op->buf.device == dev) {
+        items[n] = pending_small_pulls_[i];
+        reqs[n] = gpu::PullReq{items[n].rts, items[n].recv_op->buf.ptr, dev,
+                               items[n].size};
+        n++;
+        pending_small_pulls_.erase(pending_small_pulls_.begin() + i);
+      } else {
+        i++;
+      }
+    }
+    std::string err;
+    void* ticket = nullptr;
+    if (n == 1) {
+      ticket = gpu::begin_pull(items[0].rts, items[0].recv_op->buf,
+                               items[0].size, &err);
+    } else {
+      ticket = gpu::begin_pull_multi(reqs, n, &err);
+    }
+    if (!ticket) {
+      for (int i = 0; i < n; i++) {
+        if (items[i].conn && !items[i].conn->dead)
+          enqueue_frame(items[i].conn, FT_RECV_FAIL, 0,
+                        items[i].sender_op_id, 0, err.data(), err.size(),
+                        true);
+        fail_op(items[i].recv_op, "receive failed: " + err);
+      }
+      continue;
+    }
+    auto pull = std::make_unique<GpuPull>();
+    pull->ticket = ticket;
+    if (n == 1) {
+      pull->recv_op = items[0].recv_op;
+      pull->conn = items[0].conn;
+      pull->sender_op_id = items[0].sender_op_id;
+      pull->tag = items[0].tag;
+      pull->len = items[0].size;
+    } else {
+      pull->batch.assign(items, items + n);
+    }
+    gpu_pulls_.push_back(std::move(pull));
+  }
 }
 
 void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
@@ -1657,7 +1726,30 @@ void Engine::poll_gpu(bool& did_work) {
       continue;
     }
     did_work = true;
-    if (r > 0) {
+    if (!p->batch.empty()) {
+      for (auto& it : p->batch) {
+        if (r > 0) {
+          if (it.sender_op_id && it.conn && !it.conn->dead)
+            enqueue_frame(it.conn, FT_RECV_DONE, 0, it.sender_op_id, 0,
+                          nullptr, 0, true);
+          stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+          stats_.bytes_received.fetch_add(it.size,
+                                          std::memory_order_relaxed);
+          stats_.gpu_rx.fetch_add(1, std::memory_order_relaxed);
+          Completion comp;
+          comp.kind = Completion::Kind::RecvDone;
+          comp.op = it.recv_op;
+          comp.a = it.tag;
+          comp.b = it.size;
+          complete(std::move(comp));
+        } else {
+          if (it.sender_op_id && it.conn && !it.conn->dead)
+            enqueue_frame(it.conn, FT_RECV_FAIL, 0, it.sender_op_id, 0,
+                          err.data(), err.size(), true);
+          fail_op(it.recv_op, "receive failed: " + err);
+        }
+      }
+    } else if (r > 0) {
       if (p->sender_op_id && p->conn && !p->conn->dead)
         enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
                       true);
@@ -2111,8 +2203,10 @@ void Engine::teardown() {
     drain_commands(cmds);
     for (Op* op : cmds) fail_op(op, "operation canceled (endpoint closing)");
   }
-  // 2. Let in-flight GPU pulls finish (bounded; they are plain copies), then
-  //    complete them and best-effort ack.
+  // 2. Launch any still-pending batched pulls, then let in-flight GPU
+  //    pulls finish (bounded; they are plain copies), complete them and
+  //    best-effort ack.
+  if (!pending_small_pulls_.empty()) flush_small_pulls();
   auto deadline = std::chrono::steady_clock::now() + std::chrono::seconds(10);
   while (!gpu_pulls_.empty() &&
          std::chrono::steady_clock::now() < deadline) {

@@ -16,6 +16,13 @@
 
 namespace sw {
 // kernels.hip
+struct MultiCopyDesc {
+  const uint8_t* src;
+  uint8_t* dst;
+  uint32_t bytes;
+};
+hipError_t launch_copy_multi(const MultiCopyDesc* descs, int n,
+                             hipStream_t stream);
 hipError_t launch_copy(void* dst, const void* src, size_t bytes,
                        hipStream_t stream);
 hipError_t launch_copy_strided(void* dst, uint64_t dst_stride,
@@ -308,6 +315,48 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   hipSetDevice(prev);
   if (e != hipSuccess) {
     *err = std::string("event: ") + hipGetErrorString(e);
+    delete t;
+    return nullptr;
+  }
+  return t;
+}
+
+void* begin_pull_multi(const PullReq* reqs, int n, std::string* err) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  if (!available()) {
+    *err = "no HIP device available in receiver process";
+    return nullptr;
+  }
+  if (n < 1 || n > 8) {
+    *err = "begin_pull_multi: bad batch size";
+    return nullptr;
+  }
+  int run_dev = reqs[0].dst_device;
+  MultiCopyDesc descs[8];
+  for (int i = 0; i < n; i++) {
+    void* src = resolve_src(reqs[i].rts, run_dev, err);
+    if (!src) return nullptr;
+    bool same_proc =
+        memcmp(reqs[i].rts.src_uuid, process_uuid(), 16) == 0;
+    if (same_proc) ensure_peer_access(run_dev, reqs[i].rts.device);
+    descs[i] = {(const uint8_t*)src, reqs[i].dst_ptr,
+                (uint32_t)reqs[i].size};
+  }
+  int prev;
+  hipGetDevice(&prev);
+  hipSetDevice(run_dev);
+  hipStream_t stream = pull_stream(run_dev, 0);
+  hipError_t e = launch_copy_multi(descs, n, stream);
+  Ticket* t = nullptr;
+  if (e == hipSuccess) {
+    t = new Ticket();
+    t->device = run_dev;
+    e = pool_get_event(run_dev, &t->ev);
+    if (e == hipSuccess) e = hipEventRecord(t->ev, stream);
+  }
+  hipSetDevice(prev);
+  if (e != hipSuccess) {
+    *err = std::string("multi pull: ") + hipGetErrorString(e);
     delete t;
     return nullptr;
   }

@@ -155,6 +155,50 @@ __global__ __launch_bounds__(256) void k_copy_strided_b8(
 }
 
 // ---------------------------------------------------------------------------
+// Batched small-message copy: up to kMultiMax messages per launch, one
+// block per message (messages <= ~64 KiB). Amortizes the launch + event
+// cost that dominates small-message rate (measured: raw engine pipeline
+// sustains 600k msgs/s on host buffers but 85k with one launch+event per
+// device message).
+// ---------------------------------------------------------------------------
+struct MultiCopyDesc {
+  const uint8_t* src;
+  uint8_t* dst;
+  uint32_t bytes;
+};
+constexpr int kMultiMax = 8;
+struct MultiCopyArgs {
+  MultiCopyDesc d[kMultiMax];
+  int n;
+};
+
+__global__ __launch_bounds__(256) void k_copy_multi(MultiCopyArgs args) {
+  const MultiCopyDesc& m = args.d[blockIdx.x];
+  uintptr_t sp = (uintptr_t)m.src, dp = (uintptr_t)m.dst;
+  uint32_t bytes = m.bytes;
+  if ((sp & 15) == 0 && (dp & 15) == 0 && (bytes & 15) == 0) {
+    const uint4* s4 = (const uint4*)sp;
+    uint4* d4 = (uint4*)dp;
+    for (uint32_t i = threadIdx.x; i < bytes / 16; i += blockDim.x)
+      d4[i] = s4[i];
+  } else {
+    const uint8_t* s1 = (const uint8_t*)sp;
+    uint8_t* d1 = (uint8_t*)dp;
+    for (uint32_t i = threadIdx.x; i < bytes; i += blockDim.x)
+      d1[i] = s1[i];
+  }
+}
+
+hipError_t launch_copy_multi(const MultiCopyDesc* descs, int n,
+                             hipStream_t stream) {
+  MultiCopyArgs args;
+  for (int i = 0; i < n; i++) args.d[i] = descs[i];
+  args.n = n;
+  hipLaunchKernelGGL(k_copy_multi, dim3(n), dim3(256), 0, stream, args);
+  return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
 // Host-side dispatch
 // ---------------------------------------------------------------------------
 
