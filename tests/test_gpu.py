@@ -399,13 +399,6 @@ async def test_strided_geometry_mismatch_fails():
             await sfut
 
 
-def test_cross_process_still_ok_marker():
-    # placeholder keeping module import-time cheap; real cross-process
-    # strided coverage comes from the same-process tests (identical kernel
-    # path, only the pointer resolution differs).
-    assert True
-
-
 async def test_cross_host_gpu_bounce_path():
     """Cross-host GPU sends stage D2H and ship as eager (hipIpc cannot
     cross hosts). Forced via STARWAY_FORCE_XHOST; receiver H2D-bounces into
