@@ -419,3 +419,55 @@ async def test_cross_host_gpu_bounce_path():
             assert torch.equal(src, dst)
     finally:
         del os.environ["STARWAY_FORCE_XHOST"]
+
+def _strided_ipc_child(port, ready):
+    import torch
+
+    import starway_amd as sw
+
+    async def inner():
+        server = sw.Server()
+        server.listen("127.0.0.1", port)
+        connected = asyncio.Event()
+        loop = asyncio.get_running_loop()
+        server.set_accept_cb(lambda ep: loop.call_soon_threadsafe(connected.set))
+        ready.set()
+        await connected.wait()
+        ep = next(iter(server.list_clients()))
+        base = (torch.arange(64 * 512, dtype=torch.int32, device="cuda")
+                .reshape(64, 512) % 251).to(torch.uint8)
+        sl = base[:, 100:356]  # strided 64 x 256 slice
+        torch.cuda.synchronize()
+        await server.asend(ep, sl, 44)
+        await server.aflush_ep(ep)
+        await server.aclose()
+
+    asyncio.run(inner())
+
+
+async def test_cross_process_strided_ipc(port):
+    """Strided source over the cross-process hipIpc path: the pull kernel
+    packs rows out of the remote allocation."""
+    ctx = mp.get_context("spawn")
+    ready = ctx.Event()
+    p = ctx.Process(target=_strided_ipc_child, args=(port, ready))
+    p.start()
+    try:
+        assert ready.wait(120)
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+        dst = torch.zeros(64, 256, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        tag, ln = await client.arecv(dst, 0, 0)
+        torch.cuda.synchronize()
+        assert tag == 44 and ln == 64 * 256
+        expect = (torch.arange(64 * 512, dtype=torch.int32, device="cuda")
+                  .reshape(64, 512) % 251).to(torch.uint8)[:, 100:356]
+        assert torch.equal(dst, expect.contiguous())
+        await client.aclose()
+    finally:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
