@@ -490,7 +490,6 @@ class Engine {
   std::vector<std::pair<std::string, int>> connect_candidates_;
   py::object connect_cb_;
   bool connect_requested_ = false;
-  bool connect_done_ = false;
 
   py::object accept_cb_;
   std::mutex close_mu_;  // orders close_cb_ assignment vs teardown's read
