@@ -9,6 +9,7 @@
 //     engine's progress loop (the ucp_worker_progress analog)
 #include "core.hpp"
 
+#include <dlfcn.h>
 #include <hip/hip_runtime.h>
 
 #include <array>
@@ -46,6 +47,40 @@ static int cached_device_count() {
 
 bool available() { return cached_device_count() > 0; }
 int device_count() { return cached_device_count(); }
+
+// Optional roctx range markers (STARWAY_ROCTX=1): rocprofv3 --marker-trace
+// shows named spans around the data-plane operations. Loaded lazily via
+// dlopen so the core has no link-time roctracer dependency.
+namespace {
+typedef int (*roctx_push_fn)(const char*);
+typedef int (*roctx_pop_fn)();
+struct Roctx {
+  roctx_push_fn push = nullptr;
+  roctx_pop_fn pop = nullptr;
+  Roctx() {
+    const char* v = getenv("STARWAY_ROCTX");
+    if (!v || !strcmp(v, "0")) return;
+    void* h = dlopen("libroctx64.so.4", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) h = dlopen("libroctx64.so", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) return;
+    push = (roctx_push_fn)dlsym(h, "roctxRangePushA");
+    pop = (roctx_pop_fn)dlsym(h, "roctxRangePop");
+  }
+};
+Roctx& roctx() {
+  static Roctx r;
+  return r;
+}
+struct RoctxSpan {
+  bool active;
+  explicit RoctxSpan(const char* name) : active(roctx().push != nullptr) {
+    if (active) roctx().push(name);
+  }
+  ~RoctxSpan() {
+    if (active) roctx().pop();
+  }
+};
+}  // namespace
 
 double same_gpu_copy_gbps() { return 3100.0; }  // HBM r+w bound (measured tier)
 double xgmi_link_gbps() { return 140.0; }       // one of 7 links, sustained
@@ -120,6 +155,7 @@ using HandleKey = std::array<uint8_t, kIpcHandleBytes>;
 static std::map<std::pair<int, HandleKey>, void*> g_import_cache;
 
 bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err) {
+  RoctxSpan span("starway::make_rts");
   std::lock_guard<std::mutex> lk(g_mu);
   if (!available()) {
     *err = "no HIP device available in sender process";
@@ -229,6 +265,7 @@ static void* resolve_src(const RtsDesc& rts, int open_device,
 
 void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
                  std::string* err) {
+  RoctxSpan span("starway::pull");
   std::lock_guard<std::mutex> lk(g_mu);
   if (!available()) {
     *err = "no HIP device available in receiver process";
@@ -322,6 +359,7 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
 }
 
 void* begin_pull_multi(const PullReq* reqs, int n, std::string* err) {
+  RoctxSpan span("starway::pull_multi");
   std::lock_guard<std::mutex> lk(g_mu);
   if (!available()) {
     *err = "no HIP device available in receiver process";
