@@ -102,11 +102,23 @@ int device_of(const void* ptr) {
 // link), so streams are keyed by (device, lane) with lane derived from the
 // message source; pulls from the same peer share a lane (one link's worth
 // of bandwidth anyway, and it keeps per-pair completion work ordered).
-constexpr int kStreamLanes = 8;
+// STARWAY_LANES (default 8, power of two): more lanes overlap pulls from
+// more peers; fewer lanes reduce the per-device HSA queue count (queue
+// oversubscription across many processes sharing one GPU time-slices HW
+// queues at ~ms granularity).
+static int stream_lanes() {
+  static int lanes = [] {
+    int v = (int)env_u64("STARWAY_LANES", 8);
+    int p2 = 1;
+    while (p2 < v && p2 < 16) p2 <<= 1;
+    return p2;
+  }();
+  return lanes;
+}
 
 static hipStream_t pull_stream(int device, int lane = 0) {
   static std::map<std::pair<int, int>, hipStream_t> streams;  // g_mu held
-  lane &= (kStreamLanes - 1);
+  lane &= (stream_lanes() - 1);
   auto key = std::make_pair(device, lane);
   auto it = streams.find(key);
   if (it != streams.end()) return it->second;
@@ -282,7 +294,7 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   hipSetDevice(run_dev);
   // Lane by source peer: device ordinal + a byte of the process uuid so
   // same-device peers in different processes still spread across lanes.
-  int lane = (rts.device ^ rts.src_uuid[0]) & (kStreamLanes - 1);
+  int lane = (rts.device ^ rts.src_uuid[0]) & (stream_lanes() - 1);
   hipStream_t stream = pull_stream(run_dev, lane);
   hipError_t e;
   if (dst.device >= 0) {
@@ -530,7 +542,7 @@ void synchronize_all() {
   int prev;
   hipGetDevice(&prev);
   for (int d = 0; d < n; d++) {
-    for (int lane = 0; lane < kStreamLanes; lane++) {
+    for (int lane = 0; lane < stream_lanes(); lane++) {
       hipStream_t s = pull_stream(d, lane);
       if (s) {
         hipSetDevice(d);
