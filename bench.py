@@ -241,6 +241,17 @@ def main() -> int:
         args.msg_bytes = 16 * 1024 * 1024
     if args.chunks is None:
         args.chunks = 1 if device == "cuda" else 4
+    if device == "cuda":
+        import torch
+
+        if world > torch.cuda.device_count():
+            # Multiple ranks share a GPU (emulation): cap pull streams so
+            # the per-device HSA queue count stays under the hardware
+            # budget — oversubscribed queues are time-sliced at ~ms
+            # granularity (measured: 8 ranks x 8 lanes on one GPU ran 38x
+            # slower than lanes=1). The real 1-process-per-GPU topology
+            # keeps the default 8 lanes.
+            os.environ.setdefault("STARWAY_LANES", "1")
 
     dist = None
     if world > 1:
