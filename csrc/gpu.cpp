@@ -292,9 +292,11 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   int prev;
   hipGetDevice(&prev);
   hipSetDevice(run_dev);
-  // Lane by source peer: device ordinal + a byte of the process uuid so
-  // same-device peers in different processes still spread across lanes.
-  int lane = (rts.device ^ rts.src_uuid[0]) & (stream_lanes() - 1);
+  // Lane by SOURCE DEVICE: on a real one-process-per-GPU topology the 7
+  // peers live on 7 distinct devices => distinct lanes, full overlap
+  // across xGMI links. Same-device peers (shared-GPU emulation) share a
+  // lane, which also keeps the per-device HSA queue count low there.
+  int lane = rts.device & (stream_lanes() - 1);
   hipStream_t stream = pull_stream(run_dev, lane);
   hipError_t e;
   if (dst.device >= 0) {
