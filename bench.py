@@ -49,6 +49,11 @@ def parse_args():
                         "that the serialized same-pair kernels cannot hide "
                         "(measured); 4 on CPU where CMA/ring streaming "
                         "pipelines across chunks, +50%%)")
+    p.add_argument("--inflight", type=int, default=2,
+                   help="outstanding steps in the timed region (pipelined "
+                        "tagged rounds; 1 = strictly sequential). All K "
+                        "steps complete inside the timed window (drained "
+                        "before the closing sync/barrier).")
     p.add_argument("--lat-iters", type=int, default=200,
                    help="64B pingpong iterations for the latency probe")
     p.add_argument("--device", default=None,
@@ -119,11 +124,17 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
     chunks = max(1, min(args.chunks, per_peer))
     bounds = [per_peer * k // chunks for k in range(chunks + 1)]
     send_bufs = {j: alloc(per_peer, fill=(rank * 31 + j) % 251) for j in peers}
-    recv_bufs = {j: alloc(per_peer) for j in peers}
     send_views = {j: [send_bufs[j][bounds[k]:bounds[k + 1]]
                       for k in range(chunks)] for j in peers}
-    recv_views = {j: [recv_bufs[j][bounds[k]:bounds[k + 1]]
-                      for k in range(chunks)] for j in peers}
+    # Recv buffers are double-buffered by step parity so pipelined steps
+    # (inflight > 1) never write the same destination concurrently; send
+    # buffers are immutable after the fill, so one copy suffices.
+    recv_bufs = [{j: alloc(per_peer) for j in peers} for _ in range(2)]
+    recv_views = [
+        {j: [recv_bufs[par][j][bounds[k]:bounds[k + 1]]
+             for k in range(chunks)] for j in peers}
+        for par in range(2)
+    ]
     sync_device()
 
     mesh = None
@@ -148,32 +159,47 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
         rccl_recv = alloc(chunk * world)
         sync_device()
 
-    async def step(step_idx: int):
+    def begin_step(step_idx: int):
         if mesh is not None:
             mesh.all_to_all(rccl_send, rccl_recv)
-            mesh.synchronize()
-            return
+            return None
+        rv = recv_views[step_idx & 1]
         recvs = [
-            server.arecv(recv_views[j][k], make_tag(j, step_idx, k), full_mask)
+            server.arecv(rv[j][k], make_tag(j, step_idx, k), full_mask)
             for j in peers for k in range(chunks)
         ]
         sends = [
             clients[j].asend(send_views[j][k], make_tag(rank, step_idx, k))
             for j in peers for k in range(chunks)
         ]
-        await asyncio.gather(*sends, *recvs)
+        return asyncio.gather(*sends, *recvs)
+
+    async def run_steps(first: int, count: int, inflight: int):
+        if mesh is not None:
+            for i in range(count):
+                begin_step(first + i)
+                mesh.synchronize()
+            return
+        pending: list = []
+        for i in range(count):
+            pending.append(begin_step(first + i))
+            if len(pending) >= inflight:
+                await pending.pop(0)
+        for fut in pending:
+            await fut
+
+    inflight = max(1, args.inflight)
 
     # ---- warmup ----
-    for i in range(args.warmup):
-        await step(i)
+    await run_steps(0, args.warmup, inflight)
     sync_device()
     if world > 1:
         dist.barrier()
 
-    # ---- timed region ----
+    # ---- timed region: K pipelined steps, fully drained before the
+    # closing sync (every byte of all K steps moves inside the window) ----
     t0 = time.perf_counter()
-    for i in range(args.warmup, args.warmup + args.steps):
-        await step(i)
+    await run_steps(args.warmup, args.steps, inflight)
     sync_device()
     elapsed = time.perf_counter() - t0
     if world > 1:
@@ -291,6 +317,7 @@ def main() -> int:
                 "message_bytes_per_rank": args.msg_bytes,
                 "message_bytes_per_peer": args.msg_bytes // max(1, world - 1),
                 "chunks_per_peer": args.chunks,
+                "inflight_steps": args.inflight,
                 "endpoints": world,
                 "device": device,
                 "transport": args.transport,
