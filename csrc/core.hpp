@@ -202,6 +202,13 @@ struct Op {
   uint64_t recv_len = 0;          // actual message length for recv completion
   uint64_t recv_sender_tag = 0;
   bool gpu_send_awaiting_ack = false;
+  // Sender-side payload snapshot. Send completion means "buffer reusable":
+  // for CMA rendezvous (and eager sends whose bytes could not be written
+  // inline) the engine captures the payload here before completing, so a
+  // caller that overwrites its buffer right after `await asend` can never
+  // corrupt the in-flight message (UCX buffer-reuse contract the reference
+  // inherited from ucp_tag_send_nbx).
+  RawBuf capture;
   // Op lifetime is owned by the d2h staging list (progress_d2h reaps it);
   // connection-death cleanup must not delete it a second time.
   bool owned_by_d2h = false;
@@ -239,11 +246,16 @@ struct TxItem {
   std::vector<uint8_t> head;      // header (+ inline payload for small frames)
   const uint8_t* ext = nullptr;   // zero-copy user payload (may be null)
   size_t ext_len = 0;
-  RawBuf ext_own;                 // owned payload (cross-host GPU bounce)
+  RawBuf ext_own;                 // owned payload (captured / GPU bounce)
   bool has_keepalive = false;
   bool is_data = false;           // EAGER/RTS frames: droppable on close
   bool via_ring = false;          // carried on the shm ring, not the socket
+  bool priority = false;          // control frame inserted ahead of data
   py::object keepalive;           // dropped (under GIL) once fully written
+  // Deferred-completion owner: an eager send whose payload could not be
+  // captured (allocation failure on a huge message) completes only when its
+  // frame is fully written; dropped frames fail the op.
+  Op* owner = nullptr;
 };
 
 struct UnexpectedMsg {
@@ -256,6 +268,7 @@ struct UnexpectedMsg {
   RtsDesc rts{};
   CmaDesc cma{};
   RawBuf data;  // eager staging (uninitialized alloc)
+  uint64_t staged = 0;  // bytes counted against the conn's unexpected cap
   size_t got = 0;
   bool complete = false;
   Op* bound_recv = nullptr;   // recv matched while message still streaming in
@@ -298,6 +311,19 @@ struct Connection {
   bool shm_rx = false;      // frames are parsed from the ring
   bool sock_eof = false;    // socket closed; conn dies once the ring drains
   bool cma_denied = false;  // peer reported 'cma unavailable': send eager
+
+  // --- flow control ---
+  // Receiver side: bytes staged for unmatched (unexpected) eager messages
+  // on this connection. Above STARWAY_UNEXP_CAP the engine stops reading
+  // the connection, so TCP/ring backpressure reaches the sender instead of
+  // this process growing without bound (the UCX receive-window analog).
+  uint64_t unexp_staged_bytes = 0;
+  bool rx_paused = false;
+  // Sender side: rendezvous bytes awaiting RECV_DONE (GPU RTS / CMA / d2h)
+  // on this connection; above STARWAY_SEND_WINDOW new sends queue here and
+  // start as acks arrive (in-flight window sized for 288 GB HBM3E).
+  uint64_t inflight_rndv_bytes = 0;
+  std::deque<Op*> deferred_sends;
 
   // --- tx ---
   std::deque<TxItem> txq;
@@ -370,6 +396,8 @@ class Engine {
     std::atomic<uint64_t> bytes_sent{0}, bytes_received{0};
     std::atomic<uint64_t> eager_rx{0}, gpu_rx{0}, cma_rx{0};
     std::atomic<uint64_t> unexpected_rx{0};
+    std::atomic<uint64_t> unexp_staged_bytes{0};  // flow-control watermark
+    std::atomic<uint64_t> deferred_sends{0};      // window-queued sends
   };
   Stats stats_;
 
@@ -379,6 +407,12 @@ class Engine {
   void loop_iteration(bool& did_work);
   void drain_commands(std::vector<Op*>& cmds);
   void process_command(Op* op);
+  void start_send(Op* op, Connection* c);   // window-admitted send
+  void drain_deferred_sends(Connection* c); // window freed: start queued sends
+  void on_send_wire_handoff(Op* op, Connection* c);  // flush retarget helper
+  bool cma_eligible(const Op* op, const Connection* c) const;
+  void release_window(Op* op);              // rndv send left gpu_sends_
+  void unstage_unexp(UnexpectedMsg* um);    // unexpected staging accounting
   void do_connect_start();
   void poll_sockets(int timeout_ms, bool& did_work);
   void handle_readable(Connection* c, bool& did_work);
@@ -403,7 +437,10 @@ class Engine {
   void enqueue_frame(Connection* c, FrameType t, uint64_t tag, uint64_t op_id,
                      uint64_t aux, const void* payload, size_t payload_len,
                      bool priority);
-  void enqueue_eager(Connection* c, Op* op);
+  // Returns true when the payload is captured (written out or copied into
+  // engine-owned memory) => the op may complete immediately. False => the
+  // TxItem owns the op (deferred completion at write-out).
+  bool enqueue_eager(Connection* c, Op* op);
   void send_hello(Connection* c);
   void on_hello(Connection* c);
   void on_conn_dead(Connection* c);
@@ -480,6 +517,7 @@ class Engine {
     uint64_t size = 0;
     uint64_t done = 0;
     CmaDesc desc{};
+    RawBuf bounce;  // host staging when the posted recv buffer is on device
   };
   std::vector<std::unique_ptr<CmaPull>> cma_pulls_;
 
@@ -540,6 +578,10 @@ void* begin_h2d(const BufferRef& dst, const void* src, uint64_t size,
 // Device -> host staging download (cross-host GPU sends).
 void* begin_d2h(void* host_dst, const BufferRef& src, std::string* err);
 void attach_bounce(void* ticket, RawBuf&& bounce);
+// IPC hygiene: close every imported hipIpc mapping and drop both handle
+// caches (process-exit hook + explicit invalidation for allocators that
+// return memory to the driver, e.g. torch empty_cache).
+void ipc_close_all();
 // Poll a ticket: 1 done, 0 pending, -1 error.
 int poll_ticket(void* ticket, std::string* err);
 void free_ticket(void* ticket);

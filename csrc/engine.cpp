@@ -54,6 +54,8 @@ namespace sw {
   } while (0)
 #endif
 
+static uint64_t unexp_cap();  // defined with the matching engine below
+
 static void set_nonblocking(int fd) {
   int fl = fcntl(fd, F_GETFL, 0);
   fcntl(fd, F_SETFL, fl | O_NONBLOCK);
@@ -704,9 +706,11 @@ void Engine::loop_iteration(bool& did_work) {
   // outbound bytes are pending (the ucp_worker_progress spin analog,
   // reference main.cpp:361-468).
   // Shm rings are polled (no fd): service them every iteration.
+  const uint64_t ucap = unexp_cap();
   for (auto& c : conns_) {
     if (c->dead) continue;
-    if (c->shm_rx && c->shm && c->shm->rx.readable() > 0)
+    if (c->shm_rx && c->shm && c->shm->rx.readable() > 0 &&
+        !(ucap && c->unexp_staged_bytes >= ucap))
       handle_stream(c.get(), /*from_ring=*/true, did_work);
     if (!c->dead && !c->txq.empty() && c->txq.front().via_ring)
       handle_writable(c.get(), did_work);
@@ -721,8 +725,9 @@ void Engine::loop_iteration(bool& did_work) {
       !gpu_pulls_.empty() || !cma_pulls_.empty() || !d2h_sends_.empty();
   if (!busy)
     for (auto& c : conns_)
-      if (c->want_write() || (c->shm_rx && c->shm && !c->dead &&
-                              c->shm->rx.readable() > 0)) {
+      if (c->want_write() ||
+          (c->shm_rx && c->shm && !c->dead && c->shm->rx.readable() > 0 &&
+           !(ucap && c->unexp_staged_bytes >= ucap))) {
         busy = true;
         break;
       }
@@ -759,6 +764,65 @@ void Engine::drain_commands(std::vector<Op*>& cmds) {
   cmd_pending_.store(false, std::memory_order_release);
 }
 
+// STARWAY_SEND_WINDOW: per-connection cap on rendezvous bytes awaiting
+// RECV_DONE (GPU RTS / CMA / cross-host d2h). Sends past the window queue
+// on the connection and start as acks arrive, so a fast sender cannot pile
+// unbounded staging onto a slow receiver. Default 32 GiB — generous for
+// 288 GB HBM3E, tight enough that a runaway sender stalls long before OOM.
+static uint64_t send_window() {
+  static const uint64_t w = env_u64("STARWAY_SEND_WINDOW", 32ull << 30);
+  return w;
+}
+
+bool Engine::cma_eligible(const Op* op, const Connection* c) const {
+  // Large same-host CPU messages go rendezvous via process_vm_readv
+  // (one copy, out of band). Threshold STARWAY_CMA_THRESHOLD bytes;
+  // STARWAY_CMA=0 disables.
+  static const uint64_t cma_thresh = []() -> uint64_t {
+    const char* v = getenv("STARWAY_CMA");
+    if (v && (!strcmp(v, "0") || !strcmp(v, "false"))) return ~0ull;
+    return env_u64("STARWAY_CMA_THRESHOLD", 1 << 20);
+  }();
+  return op->buf.device < 0 && op->buf.size >= cma_thresh &&
+         !c->cma_denied && memcmp(c->peer.host_id, host_id(), 16) == 0 &&
+         memcmp(c->peer.uuid, process_uuid(), 16) != 0;
+}
+
+void Engine::release_window(Op* op) {
+  if (!op->gpu_send_awaiting_ack) return;
+  op->gpu_send_awaiting_ack = false;
+  Connection* c = op->conn;
+  if (!c) return;
+  c->inflight_rndv_bytes -=
+      std::min<uint64_t>(c->inflight_rndv_bytes, op->buf.size);
+  if (!c->dead) drain_deferred_sends(c);
+}
+
+void Engine::drain_deferred_sends(Connection* c) {
+  const uint64_t window = send_window();
+  while (!c->deferred_sends.empty() && !c->dead) {
+    Op* op = c->deferred_sends.front();
+    bool rndv = op->buf.device >= 0 || cma_eligible(op, c);
+    if (rndv && window && c->inflight_rndv_bytes > 0 &&
+        c->inflight_rndv_bytes + op->buf.size > window)
+      return;
+    c->deferred_sends.pop_front();
+    start_send(op, c);
+  }
+}
+
+void Engine::on_send_wire_handoff(Op* op, Connection* c) {
+  // A flush that snapshotted this op id (while it was windowed/staged) now
+  // waits for the wire bytes instead.
+  for (Op* f : pending_flushes_) {
+    if (f->flush_ops_pending.erase(op->id)) {
+      uint64_t target = c->tx_enqueued_bytes;
+      auto [it, ins] = f->flush_write_targets.try_emplace(c, target);
+      if (!ins && it->second < target) it->second = target;
+    }
+  }
+}
+
 void Engine::process_command(Op* op) {
   switch (op->type) {
     case OpType::Send: {
@@ -775,103 +839,18 @@ void Engine::process_command(Op* op) {
       op->conn = c;
       stats_.msgs_sent.fetch_add(1, std::memory_order_relaxed);
       stats_.bytes_sent.fetch_add(op->buf.size, std::memory_order_relaxed);
-      // Large same-host CPU messages go rendezvous via process_vm_readv
-      // (one copy, out of band). Threshold STARWAY_CMA_THRESHOLD bytes;
-      // STARWAY_CMA=0 disables. Send completes with eager semantics
-      // (handed off) — delivery is still only guaranteed by flush, which
-      // waits for the RECV_DONE of pending CMA ops.
-      static const uint64_t cma_thresh = []() -> uint64_t {
-        const char* v = getenv("STARWAY_CMA");
-        if (v && (!strcmp(v, "0") || !strcmp(v, "false")))
-          return ~0ull;  // disabled
-        return env_u64("STARWAY_CMA_THRESHOLD", 1 << 20);
-      }();
-      if (op->buf.device < 0 && op->buf.size >= cma_thresh &&
-          !c->cma_denied &&
-          memcmp(c->peer.host_id, host_id(), 16) == 0 &&
-          memcmp(c->peer.uuid, process_uuid(), 16) != 0) {
-        CmaDesc desc{};
-        desc.pid = (uint64_t)getpid();
-        desc.addr = (uint64_t)(uintptr_t)op->buf.ptr;
-        memcpy(desc.src_uuid, process_uuid(), 16);
-        enqueue_frame(c, FT_RTS_CPU, op->tag, op->id, op->buf.size, &desc,
-                      sizeof(desc), /*priority=*/false);
-        op->gpu_send_awaiting_ack = true;  // same ack machinery as GPU RTS
-        gpu_sends_[op->id] = op;
-        // Eager-style completion: buffer handed off; keepalive stays
-        // pinned in the op until RECV_DONE (or cancel) so the receiver's
-        // pull reads live memory. Both callbacks are consumed here — a
-        // later cancel must not fire fail_cb on an already-completed op.
-        {
-          py::gil_scoped_acquire gil;
-          try {
-            if (op->done_cb.ptr()) op->done_cb();
-          } catch (py::error_already_set& e) {
-            e.discard_as_unraisable("starway send callback");
-          }
-          op->done_cb = py::object();
-          op->fail_cb = py::object();
-        }
+      // Window admission. Any send behind already-deferred ones must also
+      // queue, preserving per-connection message order for tag matching.
+      const uint64_t window = send_window();
+      bool rndv = op->buf.device >= 0 || cma_eligible(op, c);
+      if (!c->deferred_sends.empty() ||
+          (rndv && window && c->inflight_rndv_bytes > 0 &&
+           c->inflight_rndv_bytes + op->buf.size > window)) {
+        c->deferred_sends.push_back(op);
+        stats_.deferred_sends.fetch_add(1, std::memory_order_relaxed);
         return;
       }
-      static const bool force_xhost =
-          getenv("STARWAY_FORCE_XHOST") != nullptr;  // test hook
-      if (op->buf.device >= 0 &&
-          (force_xhost ||
-           memcmp(c->peer.host_id, host_id(), 16) != 0)) {
-        // Cross-host GPU send: hipIpc cannot cross hosts — stage the
-        // payload to host memory and ship it as a plain eager frame once
-        // the download completes. The op completes at hand-off (eager
-        // semantics); both callbacks are consumed here.
-        auto d2h = std::make_unique<D2hSend>();
-        if (!d2h->buf.alloc(op->buf.size)) {
-          fail_op(op, "send failed: staging allocation failed");
-          return;
-        }
-        std::string err;
-        d2h->ticket = gpu::begin_d2h(d2h->buf.data(), op->buf, &err);
-        if (!d2h->ticket) {
-          fail_op(op, "send failed: " + err);
-          return;
-        }
-        d2h->op = op;
-        {
-          py::gil_scoped_acquire gil;
-          try {
-            if (op->done_cb.ptr()) op->done_cb();
-          } catch (py::error_already_set& e) {
-            e.discard_as_unraisable("starway send callback");
-          }
-          op->done_cb = py::object();
-          op->fail_cb = py::object();
-        }
-        // Track it like a GPU send so flush waits for the wire hand-off.
-        gpu_sends_[op->id] = op;
-        op->gpu_send_awaiting_ack = true;
-        op->owned_by_d2h = true;  // progress_d2h reaps; see on_conn_dead
-        d2h_sends_.push_back(std::move(d2h));
-        return;
-      }
-      if (op->buf.device >= 0) {
-        // GPU rendezvous.
-        RtsDesc rts{};
-        std::string err;
-        if (!gpu::make_rts(op->buf, &rts, &err)) {
-          fail_op(op, "send failed: " + err);
-          return;
-        }
-        enqueue_frame(c, FT_RTS, op->tag, op->id, op->buf.size, &rts,
-                      sizeof(rts), /*priority=*/false);
-        op->gpu_send_awaiting_ack = true;
-        gpu_sends_[op->id] = op;
-      } else {
-        enqueue_eager(c, op);
-        // CPU send: handed to transport => complete (eager semantics).
-        Completion comp;
-        comp.kind = Completion::Kind::SendDone;
-        comp.op = op;
-        complete(std::move(comp));
-      }
+      start_send(op, c);
       break;
     }
     case OpType::Recv:
@@ -900,6 +879,10 @@ void Engine::process_command(Op* op) {
             (!targets.empty() && sop->conn == targets[0]))
           op->flush_ops_pending.insert(id);
       }
+      // Window-deferred sends were posted before this flush: cover them too
+      // (their ids migrate to gpu_sends_ / write targets when they start).
+      for (Connection* c : targets)
+        for (Op* d : c->deferred_sends) op->flush_ops_pending.insert(d->id);
       if (op->flush_write_targets.empty() && op->flush_ops_pending.empty()) {
         Completion comp;
         comp.kind = Completion::Kind::FlushDone;
@@ -915,6 +898,103 @@ void Engine::process_command(Op* op) {
   }
 }
 
+// One window-admitted send. Completion contract (reference flush tests,
+// tests/test_basic.py:250-416 + UCX buffer-reuse semantics): the done
+// callback means "the engine owns the payload" — either the bytes are on
+// the wire, captured into engine memory, or (GPU rendezvous) delivery is
+// tracked to RECV_DONE. Delivery is only guaranteed after flush.
+void Engine::start_send(Op* op, Connection* c) {
+  if (cma_eligible(op, c)) {
+    CmaDesc desc{};
+    desc.pid = (uint64_t)getpid();
+    memcpy(desc.src_uuid, process_uuid(), 16);
+    // Capture: snapshot the payload so the caller may overwrite its buffer
+    // the moment `await asend` returns; the receiver pulls the snapshot.
+    // On allocation failure fall back to zero-copy from the live buffer
+    // (keepalive pins it; contents then must not change until flush).
+    if (op->capture.alloc(op->buf.size)) {
+      memcpy(op->capture.data(), op->buf.ptr, op->buf.size);
+      desc.addr = (uint64_t)(uintptr_t)op->capture.data();
+      dead_objs_.push_back(std::move(op->keepalive));
+    } else {
+      desc.addr = (uint64_t)(uintptr_t)op->buf.ptr;
+    }
+    enqueue_frame(c, FT_RTS_CPU, op->tag, op->id, op->buf.size, &desc,
+                  sizeof(desc), /*priority=*/false);
+    op->gpu_send_awaiting_ack = true;  // same ack machinery as GPU RTS
+    gpu_sends_[op->id] = op;
+    c->inflight_rndv_bytes += op->buf.size;
+    // Eager-style completion: payload captured/pinned; delivery guaranteed
+    // only by flush (waits for RECV_DONE). Both callbacks are consumed so
+    // a later cancel cannot fire fail_cb on a completed op.
+    {
+      py::gil_scoped_acquire gil;
+      try {
+        if (op->done_cb.ptr()) op->done_cb();
+      } catch (py::error_already_set& e) {
+        e.discard_as_unraisable("starway send callback");
+      }
+      op->done_cb = py::object();
+      op->fail_cb = py::object();
+    }
+    return;
+  }
+  static const bool force_xhost =
+      getenv("STARWAY_FORCE_XHOST") != nullptr;  // test hook
+  if (op->buf.device >= 0 &&
+      (force_xhost || memcmp(c->peer.host_id, host_id(), 16) != 0)) {
+    // Cross-host GPU send: hipIpc cannot cross hosts — stage the payload
+    // to host memory and ship it as a plain eager frame once the download
+    // completes. The op completes when the d2h ticket lands (payload
+    // captured in the bounce => buffer reusable), in progress_d2h.
+    auto d2h = std::make_unique<D2hSend>();
+    if (!d2h->buf.alloc(op->buf.size)) {
+      fail_op(op, "send failed: staging allocation failed");
+      return;
+    }
+    std::string err;
+    d2h->ticket = gpu::begin_d2h(d2h->buf.data(), op->buf, &err);
+    if (!d2h->ticket) {
+      fail_op(op, "send failed: " + err);
+      return;
+    }
+    d2h->op = op;
+    // Track it like a GPU send so flush waits for the wire hand-off.
+    gpu_sends_[op->id] = op;
+    op->gpu_send_awaiting_ack = true;
+    c->inflight_rndv_bytes += op->buf.size;
+    op->owned_by_d2h = true;  // progress_d2h reaps; see on_conn_dead
+    d2h_sends_.push_back(std::move(d2h));
+    return;
+  }
+  if (op->buf.device >= 0) {
+    // GPU rendezvous: RTS -> receiver pulls over xGMI -> RECV_DONE. The
+    // send completes at RECV_DONE (delivery), so buffer reuse is safe.
+    RtsDesc rts{};
+    std::string err;
+    if (!gpu::make_rts(op->buf, &rts, &err)) {
+      fail_op(op, "send failed: " + err);
+      return;
+    }
+    enqueue_frame(c, FT_RTS, op->tag, op->id, op->buf.size, &rts,
+                  sizeof(rts), /*priority=*/false);
+    op->gpu_send_awaiting_ack = true;
+    gpu_sends_[op->id] = op;
+    c->inflight_rndv_bytes += op->buf.size;
+  } else {
+    bool captured = enqueue_eager(c, op);
+    on_send_wire_handoff(op, c);  // windowed op: flush id -> write target
+    if (captured) {
+      // Payload written or captured => complete now (eager semantics).
+      Completion comp;
+      comp.kind = Completion::Kind::SendDone;
+      comp.op = op;
+      complete(std::move(comp));
+    }
+    // else: the TxItem owns the op; completion fires at write-out.
+  }
+}
+
 // ---- socket IO ------------------------------------------------------------
 
 void Engine::poll_sockets(int timeout_ms, bool& did_work) {
@@ -926,11 +1006,16 @@ void Engine::poll_sockets(int timeout_ms, bool& did_work) {
     pfds.push_back({listen_fd_, POLLIN, 0});
   }
   size_t conn_base = pfds.size();
+  const uint64_t ucap = unexp_cap();
   std::vector<Connection*> live;
   for (auto& c : conns_) {
     if (c->dead || c->fd < 0) continue;
-    short ev = POLLIN;
+    // Receive window: over the unexpected-staging cap, stop reading this
+    // connection (socket backpressure) but keep writing our own frames.
+    short ev =
+        (ucap && c->unexp_staged_bytes >= ucap) ? 0 : (short)POLLIN;
     if (c->want_write()) ev |= POLLOUT;
+    if (!ev) continue;
     pfds.push_back({c->fd, ev, 0});
     live.push_back(c.get());
   }
@@ -1025,10 +1110,17 @@ void Engine::handle_stream(Connection* c, bool from_ring, bool& did_work) {
     if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return -2;
     return n;
   };
+  const uint64_t ucap = unexp_cap();
   while (!c->dead) {
     // A mid-stream rx-source switch (SHM_SWITCH parsed from TCP) hands the
     // remaining frames to the ring reader invoked from the engine loop.
     if (c->shm_rx != from_ring) return;
+    // Receive window: stop consuming at the next frame boundary once the
+    // unexpected backlog crosses the cap (bounds overshoot to one message;
+    // parser state is preserved and resumes when recvs drain the backlog).
+    if (ucap && c->unexp_staged_bytes >= ucap &&
+        c->rx_state == Connection::RxState::Header && c->rx_got == 0)
+      return;
     if (c->rx_state == Connection::RxState::Header) {
       uint8_t* hp = (uint8_t*)&c->rx_hdr;
       ssize_t n = src_read(hp + c->rx_got, sizeof(FrameHeader) - c->rx_got);
@@ -1382,6 +1474,13 @@ void Engine::begin_eager(Connection* c) {
       on_conn_dead(c);
       return;
     }
+    // Flow control: staged bytes count against the connection's receive
+    // window; past STARWAY_UNEXP_CAP the engine stops reading this
+    // connection until recvs drain the backlog (TCP/ring backpressure
+    // reaches the sender instead of unbounded staging growth).
+    um->staged = msg_len;
+    c->unexp_staged_bytes += msg_len;
+    stats_.unexp_staged_bytes.fetch_add(msg_len, std::memory_order_relaxed);
     c->rx_unexp = um.get();
     unexpected_.push_back(std::move(um));
   }
@@ -1575,7 +1674,8 @@ void Engine::flush_small_pulls() {
 void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
                             uint64_t size, uint64_t sender_op,
                             Connection* c) {
-  if (size > recv_op->buf.size) {
+  if (size > recv_op->buf.size ||
+      (recv_op->buf.rows > 0 && size != recv_op->buf.size)) {
     std::string err = "message truncated (len " + std::to_string(size) +
                       " > buffer " + std::to_string(recv_op->buf.size) + ")";
     enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
@@ -1584,6 +1684,19 @@ void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
     return;
   }
   auto pull = std::make_unique<CmaPull>();
+  if (recv_op->buf.device >= 0) {
+    // The posted recv buffer lives on a GPU: process_vm_readv cannot write
+    // device memory, so pull into a host bounce and upload with begin_h2d
+    // at completion. If the bounce cannot be allocated, decline CMA — the
+    // sender retransmits as eager, which handles device recvs natively.
+    if (!pull->bounce.alloc(size)) {
+      std::string err = "cma unavailable (bounce allocation failed)";
+      enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
+                    true);
+      posted_recvs_.push_front(recv_op);
+      return;
+    }
+  }
   pull->recv_op = recv_op;
   pull->conn = c;
   pull->sender_op_id = sender_op;
@@ -1605,6 +1718,8 @@ void Engine::progress_d2h(bool& did_work) {
     did_work = true;
     Op* op = p->op;
     gpu::free_ticket(p->ticket);
+    gpu_sends_.erase(op->id);
+    release_window(op);
     if (r > 0 && op->conn && !op->conn->dead) {
       Connection* c = op->conn;
       TxItem item;
@@ -1626,22 +1741,18 @@ void Engine::progress_d2h(bool& did_work) {
       c->txq.push_back(std::move(item));
       bool dummy = false;
       handle_writable(c, dummy);
-      // Flushes that snapshot this op id now wait for the wire bytes.
-      for (Op* f : pending_flushes_) {
-        if (f->flush_ops_pending.erase(op->id)) {
-          uint64_t target = c->tx_enqueued_bytes;
-          auto [fit, ins] = f->flush_write_targets.try_emplace(c, target);
-          if (!ins && fit->second < target) fit->second = target;
-        }
-      }
+      // Flushes that snapshot this op id now wait for the wire bytes; the
+      // payload is captured in the frame => the send completes here.
+      on_send_wire_handoff(op, c);
+      Completion comp;
+      comp.kind = Completion::Kind::SendDone;
+      comp.op = op;
+      complete(std::move(comp));
     } else {
       for (Op* f : pending_flushes_) f->flush_ops_pending.erase(op->id);
+      fail_op(op, r > 0 ? "send failed: connection reset"
+                        : "send failed: " + err);
     }
-    gpu_sends_.erase(op->id);
-    Completion comp;  // callbacks already consumed; just reaps the op
-    comp.kind = Completion::Kind::SendDone;
-    comp.op = op;
-    complete(std::move(comp));
     d2h_sends_.erase(d2h_sends_.begin() + i);
   }
 }
@@ -1660,7 +1771,9 @@ void Engine::progress_cma(bool& did_work) {
       continue;
     }
     size_t want = (size_t)std::min<uint64_t>(kChunk, p->size - p->done);
-    struct iovec liov {p->recv_op->buf.ptr + p->done, want};
+    uint8_t* dst_base =
+        p->bounce.size() ? p->bounce.data() : p->recv_op->buf.ptr;
+    struct iovec liov {dst_base + p->done, want};
     struct iovec riov {(void*)(uintptr_t)(p->desc.addr + p->done), want};
     ssize_t n;
     static const bool force_eperm =
@@ -1698,11 +1811,35 @@ void Engine::progress_cma(bool& did_work) {
     did_work = true;
     p->done += (uint64_t)n;
     if (p->done >= p->size) {
+      // Data is out of the sender's address space => ack now (the sender
+      // may reuse/free its buffer); any device upload is local work.
       enqueue_frame(p->conn, FT_RECV_DONE, 0, p->sender_op_id, 0, nullptr, 0,
                     true);
+      stats_.cma_rx.fetch_add(1, std::memory_order_relaxed);
+      if (p->bounce.size()) {
+        // Device recv: upload the bounce; the recv completes with the h2d
+        // ticket (poll_gpu), which also owns the bounce's lifetime.
+        std::string err;
+        void* ticket =
+            gpu::begin_h2d(p->recv_op->buf, p->bounce.data(), p->size, &err);
+        if (!ticket) {
+          fail_op(p->recv_op, "receive failed: " + err);
+        } else {
+          auto gp = std::make_unique<GpuPull>();
+          gp->ticket = ticket;
+          gp->recv_op = p->recv_op;
+          gp->conn = nullptr;
+          gp->sender_op_id = 0;
+          gp->tag = p->tag;
+          gp->len = p->size;
+          gpu::attach_bounce(ticket, std::move(p->bounce));
+          gpu_pulls_.push_back(std::move(gp));
+        }
+        cma_pulls_.erase(cma_pulls_.begin() + i);
+        continue;
+      }
       stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
       stats_.bytes_received.fetch_add(p->size, std::memory_order_relaxed);
-      stats_.cma_rx.fetch_add(1, std::memory_order_relaxed);
       Completion comp;
       comp.kind = Completion::Kind::RecvDone;
       comp.op = p->recv_op;
@@ -1779,27 +1916,26 @@ void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
   if (it == gpu_sends_.end()) return;
   Op* op = it->second;
   gpu_sends_.erase(it);
+  release_window(op);
   if (failed && op->buf.device < 0 &&
       err.find("cma unavailable") != std::string::npos && op->conn &&
       !op->conn->dead) {
     op->conn->cma_denied = true;  // stop offering CMA on this connection
     // Receiver cannot process_vm_readv us (e.g. yama ptrace restrictions):
-    // retransmit the message as a plain eager stream. The send op already
+    // retransmit the message as a plain eager stream (the captured
+    // snapshot, when present, becomes the frame body). The send op already
     // completed (handed off); extend any pending flush that covered it to
     // the new write target so flush still means delivery.
-    enqueue_eager(op->conn, op);
-    for (Op* f : pending_flushes_) {
-      if (f->flush_ops_pending.erase(op_id)) {
-        uint64_t target = op->conn->tx_enqueued_bytes;
-        auto [fit, inserted] = f->flush_write_targets.try_emplace(
-            op->conn, target);
-        if (!inserted && fit->second < target) fit->second = target;
-      }
+    bool captured = enqueue_eager(op->conn, op);
+    on_send_wire_handoff(op, op->conn);
+    if (captured) {
+      Completion comp;  // silent completion: callbacks already consumed
+      comp.kind = Completion::Kind::SendDone;
+      comp.op = op;
+      complete(std::move(comp));
     }
-    Completion comp;  // silent completion: callbacks already consumed
-    comp.kind = Completion::Kind::SendDone;
-    comp.op = op;
-    complete(std::move(comp));
+    // else: the TxItem owns the op and reaps it at write-out (its
+    // callbacks are already consumed, so the completion is silent there).
     return;
   }
   if (failed) {
@@ -1827,6 +1963,21 @@ void Engine::on_gpu_send_acked(uint64_t op_id, bool failed,
 }
 
 // ---- recv matching --------------------------------------------------------
+
+// STARWAY_UNEXP_CAP: per-connection unexpected-staging receive window
+// (0 disables). Chosen well above the test suite's 32 MiB staging cases
+// but far below RAM scale.
+static uint64_t unexp_cap() {
+  static const uint64_t cap = env_u64("STARWAY_UNEXP_CAP", 512ull << 20);
+  return cap;
+}
+
+void Engine::unstage_unexp(UnexpectedMsg* um) {
+  if (!um->staged) return;
+  if (um->conn) um->conn->unexp_staged_bytes -= um->staged;
+  stats_.unexp_staged_bytes.fetch_sub(um->staged, std::memory_order_relaxed);
+  um->staged = 0;
+}
 
 void Engine::match_or_stash_recv(Op* op) {
   if (!try_match_unexpected(op)) posted_recvs_.push_back(op);
@@ -1860,6 +2011,7 @@ bool Engine::try_match_unexpected(Op* op) {
         memcpy(op->buf.ptr, um->data.data(), um->got);
         um->redirect_dst = op->buf.ptr;
         um->data.clear();
+        unstage_unexp(um);
       }
       return true;
     }
@@ -1873,6 +2025,7 @@ bool Engine::try_match_unexpected(Op* op) {
 
 void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
   std::unique_ptr<UnexpectedMsg> guard(um);
+  unstage_unexp(um);
   if (um->redirect_dst) {
     // Stream was redirected into the recv buffer; payload already in place.
     Completion comp;
@@ -1938,10 +2091,15 @@ void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
   if (payload_len) memcpy(item.head.data() + sizeof(h), payload, payload_len);
   item.is_data = (t == FT_RTS);
   item.via_ring = c->shm_tx_enq;
+  item.priority = priority;
   c->tx_enqueued_bytes += item.head.size();
   if (priority && !c->txq.empty()) {
-    // Insert at the first frame boundary, after any queued priority frames.
+    // Insert at a frame boundary: past the partially-written front frame
+    // and past every already-queued priority frame, so priority control
+    // frames keep their relative order and never jump a queued handshake
+    // marker (SHM_ACK/SHM_SWITCH "last TCP frame" invariant).
     size_t pos = c->tx_front_written > 0 ? 1 : 0;
+    while (pos < c->txq.size() && c->txq[pos].priority) pos++;
     c->txq.insert(c->txq.begin() + pos, std::move(item));
   } else {
     c->txq.push_back(std::move(item));
@@ -1950,7 +2108,7 @@ void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
   handle_writable(c, dummy);  // opportunistic immediate write
 }
 
-void Engine::enqueue_eager(Connection* c, Op* op) {
+bool Engine::enqueue_eager(Connection* c, Op* op) {
   TxItem item;
   static const size_t kInline = env_u64("STARWAY_EAGER_INLINE", 4096);
   FrameHeader h{};
@@ -1962,7 +2120,15 @@ void Engine::enqueue_eager(Connection* c, Op* op) {
   h.aux = op->buf.size;
   item.is_data = true;
   item.via_ring = c->shm_tx_enq;
-  if (op->buf.size <= kInline) {
+  if (op->capture.size()) {
+    // Retransmission of an already-captured payload (CMA fallback): the
+    // snapshot becomes the frame body, zero extra copies.
+    item.head.resize(sizeof(h));
+    memcpy(item.head.data(), &h, sizeof(h));
+    item.ext_own = std::move(op->capture);
+    item.ext = item.ext_own.data();
+    item.ext_len = op->buf.size;
+  } else if (op->buf.size <= kInline) {
     item.head.resize(sizeof(h) + op->buf.size);
     memcpy(item.head.data(), &h, sizeof(h));
     memcpy(item.head.data() + sizeof(h), op->buf.ptr, op->buf.size);
@@ -1978,6 +2144,32 @@ void Engine::enqueue_eager(Connection* c, Op* op) {
   c->txq.push_back(std::move(item));
   bool dummy = false;
   handle_writable(c, dummy);
+  // Capture-on-completion: if the zero-copy frame was not fully written by
+  // the opportunistic write above, snapshot the unwritten remainder into
+  // engine-owned memory before the send completes, so no live pointer into
+  // the caller's buffer survives completion (buffer-reuse contract).
+  if (!c->txq.empty()) {
+    TxItem& back = c->txq.back();
+    if (back.ext && back.ext_own.empty() && back.ext == op->buf.ptr) {
+      size_t off = 0;
+      if (&back == &c->txq.front() && c->tx_front_written > back.head.size())
+        off = c->tx_front_written - back.head.size();
+      if (back.ext_own.alloc(back.ext_len)) {
+        memcpy(back.ext_own.data() + off, back.ext + off, back.ext_len - off);
+        back.ext = back.ext_own.data();
+        if (back.has_keepalive) {
+          dead_objs_.push_back(std::move(back.keepalive));
+          back.has_keepalive = false;
+        }
+      } else {
+        // Snapshot allocation failed (enormous payload under memory
+        // pressure): keep zero-copy and defer completion to write-out.
+        back.owner = op;
+        return false;
+      }
+    }
+  }
+  return true;
 }
 
 void Engine::handle_writable(Connection* c, bool& did_work) {
@@ -2025,6 +2217,14 @@ void Engine::handle_writable(Connection* c, bool& did_work) {
       if (it.has_keepalive) {
         dead_objs_.push_back(std::move(it.keepalive));
         it.has_keepalive = false;
+      }
+      if (it.owner) {
+        // Deferred-completion eager send: payload now fully on the wire.
+        Completion comp;
+        comp.kind = Completion::Kind::SendDone;
+        comp.op = it.owner;
+        it.owner = nullptr;
+        complete(std::move(comp));
       }
       c->txq.pop_front();
       c->tx_front_written = 0;
@@ -2081,17 +2281,28 @@ void Engine::on_conn_dead(Connection* c) {
     for (auto it = unexpected_.begin(); it != unexpected_.end(); ++it) {
       if (it->get() == c->rx_unexp) {
         if ((*it)->bound_recv) posted_recvs_.push_front((*it)->bound_recv);
+        unstage_unexp(it->get());
         unexpected_.erase(it);
         break;
       }
     }
     c->rx_unexp = nullptr;
   }
-  // Drop queued tx (deferring py keepalive refs).
+  // Complete staged unexpected messages from this conn stay matchable
+  // (data fully arrived before death), but stop counting against the
+  // receive window of a connection that no longer reads.
+  for (auto& um : unexpected_)
+    if (um->conn == c) unstage_unexp(um.get());
+  // Drop queued tx (deferring py keepalive refs; failing deferred-
+  // completion owners — their payload never made the wire).
   for (auto& item : c->txq) {
     if (item.has_keepalive) {
       dead_objs_.push_back(std::move(item.keepalive));
       item.has_keepalive = false;
+    }
+    if (item.owner) {
+      fail_op(item.owner, "send failed: connection reset");
+      item.owner = nullptr;
     }
   }
   c->txq.clear();
@@ -2124,14 +2335,39 @@ void Engine::on_conn_dead(Connection* c) {
   }
   // GPU sends routed to this conn will never be acked. Ops owned by the
   // d2h staging list are only unregistered here — progress_d2h reaps them
-  // (double-delete hazard otherwise).
+  // (double-delete hazard otherwise). Collect the op ids so flushes that
+  // snapshotted them fail instead of hanging forever.
+  std::vector<uint64_t> dead_send_ids;
   for (auto it = gpu_sends_.begin(); it != gpu_sends_.end();) {
     if (it->second->conn == c) {
+      dead_send_ids.push_back(it->first);
+      it->second->gpu_send_awaiting_ack = false;  // window already dying
       if (!it->second->owned_by_d2h)
         fail_op(it->second, "send failed: connection reset");
       it = gpu_sends_.erase(it);
     } else {
       ++it;
+    }
+  }
+  // Window-deferred sends never started: fail them too.
+  for (Op* d : c->deferred_sends) {
+    dead_send_ids.push_back(d->id);
+    fail_op(d, "send failed: connection reset");
+  }
+  c->deferred_sends.clear();
+  c->inflight_rndv_bytes = 0;
+  if (!dead_send_ids.empty()) {
+    for (size_t i = 0; i < pending_flushes_.size();) {
+      Op* f = pending_flushes_[i];
+      bool hit = false;
+      for (uint64_t id : dead_send_ids)
+        if (f->flush_ops_pending.erase(id)) hit = true;
+      if (hit) {
+        fail_op(f, "flush failed: connection reset");
+        pending_flushes_.erase(pending_flushes_.begin() + i);
+      } else {
+        i++;
+      }
     }
   }
   if (mode_ == ClientMode && status_.load() == 2) {
@@ -2240,10 +2476,9 @@ void Engine::teardown() {
   for (auto& p : d2h_sends_) {
     gpu::free_ticket(p->ticket);
     gpu_sends_.erase(p->op->id);
-    Completion comp;  // callbacks consumed at hand-off; reap silently
-    comp.kind = Completion::Kind::SendDone;
-    comp.op = p->op;
-    complete(std::move(comp));
+    // Never completed (completion now waits for the d2h ticket) and the
+    // endpoint is closing before the payload reached the wire: cancel.
+    fail_op(p->op, "operation canceled (endpoint closing)");
   }
   d2h_sends_.clear();
   // 3. Cancel in-flight data: a connection with undelivered EAGER/RTS bytes
@@ -2270,10 +2505,17 @@ void Engine::teardown() {
   //    main.cpp:680-701, tests/test_basic.py:638-663).
   for (Op* r : posted_recvs_) fail_op(r, "operation canceled (endpoint closing)");
   posted_recvs_.clear();
-  for (auto& um : unexpected_)
+  for (auto& um : unexpected_) {
+    unstage_unexp(um.get());
     if (um->bound_recv)
       fail_op(um->bound_recv, "operation canceled (endpoint closing)");
+  }
   unexpected_.clear();
+  for (auto& c : conns_) {
+    for (Op* d : c->deferred_sends)
+      fail_op(d, "operation canceled (endpoint closing)");
+    c->deferred_sends.clear();
+  }
   for (Op* f : pending_flushes_) fail_op(f, "operation canceled (endpoint closing)");
   pending_flushes_.clear();
   for (auto& [id, op] : gpu_sends_)
@@ -2301,11 +2543,16 @@ void Engine::teardown() {
       c->fd = -1;
     }
     if (c->ep) c->ep->conn = nullptr;
-    for (auto& item : c->txq)
+    for (auto& item : c->txq) {
       if (item.has_keepalive) {
         dead_objs_.push_back(std::move(item.keepalive));
         item.has_keepalive = false;
       }
+      if (item.owner) {
+        fail_op(item.owner, "operation canceled (endpoint closing)");
+        item.owner = nullptr;
+      }
+    }
     c->txq.clear();
     c->shm_rx = false;
     c->shm.reset();

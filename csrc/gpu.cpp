@@ -14,6 +14,7 @@
 
 #include <array>
 #include <cstring>
+#include <tuple>
 
 namespace sw {
 // kernels.hip
@@ -155,16 +156,36 @@ static void ensure_peer_access(int dst_dev, int src_dev) {
 // IPC handle caches
 // ---------------------------------------------------------------------------
 
-// Export cache: base ptr -> handle (registration-cache pattern; torch's
-// caching allocator keeps blocks alive so base pointers stay valid).
+// Export cache: base ptr -> (handle, allocation size). Registration-cache
+// pattern: torch's caching allocator keeps blocks alive, so base pointers
+// stay valid across messages. Entries are VALIDATED on lookup by comparing
+// the allocation size hipMemGetAddressRange reports — a freed-and-reused
+// base with a different size re-exports. A reuse at the same base AND the
+// same size cannot be detected this way; callers using allocators that
+// return memory to the driver mid-run (e.g. torch empty_cache) must call
+// starway_amd.ipc_invalidate() (-> ipc_close_all) at that point.
 struct ExportEntry {
   hipIpcMemHandle_t handle;
+  size_t size = 0;
 };
 static std::map<void*, ExportEntry> g_export_cache;
 
 // Import cache: (device, 64B handle) -> mapped base.
 using HandleKey = std::array<uint8_t, kIpcHandleBytes>;
 static std::map<std::pair<int, HandleKey>, void*> g_import_cache;
+
+void ipc_close_all() {
+  std::lock_guard<std::mutex> lk(g_mu);
+  int prev = 0;
+  hipGetDevice(&prev);
+  for (auto& [key, base] : g_import_cache) {
+    hipSetDevice(key.first);
+    hipIpcCloseMemHandle(base);
+  }
+  hipSetDevice(prev);
+  g_import_cache.clear();
+  g_export_cache.clear();
+}
 
 bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err) {
   RoctxSpan span("starway::make_rts");
@@ -189,8 +210,14 @@ bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err) {
   hipError_t e = hipMemGetAddressRange(&base, &bsize, (hipDeviceptr_t)buf.ptr);
   if (e == hipSuccess) {
     auto it = g_export_cache.find((void*)base);
+    if (it != g_export_cache.end() && it->second.size != bsize) {
+      // Base address reused by a different allocation: stale entry.
+      g_export_cache.erase(it);
+      it = g_export_cache.end();
+    }
     if (it == g_export_cache.end()) {
       ExportEntry ent;
+      ent.size = bsize;
       e = hipIpcGetMemHandle(&ent.handle, (void*)base);
       if (e == hipSuccess) {
         it = g_export_cache.emplace((void*)base, ent).first;
@@ -275,6 +302,26 @@ static void* resolve_src(const RtsDesc& rts, int open_device,
   return (uint8_t*)base + rts.offset;
 }
 
+// First-contact route self-check: one stderr line per (dst device, src
+// device, process locality) triple, so a multi-GPU driver-run failure is
+// diagnosable from the log tail (which xGMI pair, ipc vs raw pointer,
+// which lane, which copy engine). STARWAY_LOG_ROUTE=0 silences it.
+static void log_route_once(int run_dev, const RtsDesc& rts, bool same_proc,
+                           int lane, const char* engine) {
+  static const bool enabled = [] {
+    const char* v = getenv("STARWAY_LOG_ROUTE");
+    return !(v && !strcmp(v, "0"));
+  }();
+  if (!enabled) return;
+  static std::set<std::tuple<int, int, int>> seen;  // g_mu held by callers
+  if (!seen.insert({run_dev, rts.device, (int)same_proc}).second) return;
+  fprintf(stderr,
+          "[starway] route: pull dev%d <- peer dev%d (%s, %s) lane=%d "
+          "engine=%s\n",
+          run_dev, rts.device, same_proc ? "same-proc" : "cross-proc",
+          rts.use_ipc ? "hipipc" : "raw-ptr", lane, engine);
+}
+
 void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
                  std::string* err) {
   RoctxSpan span("starway::pull");
@@ -302,6 +349,7 @@ void* begin_pull(const RtsDesc& rts, const BufferRef& dst, uint64_t size,
   if (dst.device >= 0) {
     bool same_proc = memcmp(rts.src_uuid, process_uuid(), 16) == 0;
     if (same_proc) ensure_peer_access(dst.device, rts.device);
+    log_route_once(run_dev, rts, same_proc, lane, "gfx950_copy");
     // Pull engine selection: gfx950 copy kernel (default) or the SDMA
     // engines via hipMemcpyPeerAsync/hipMemcpyAsync (STARWAY_PULL=sdma).
     // SDMA leaves CUs free and can ride dedicated copy engines; the kernel
@@ -391,6 +439,7 @@ void* begin_pull_multi(const PullReq* reqs, int n, std::string* err) {
     bool same_proc =
         memcmp(reqs[i].rts.src_uuid, process_uuid(), 16) == 0;
     if (same_proc) ensure_peer_access(run_dev, reqs[i].rts.device);
+    log_route_once(run_dev, reqs[i].rts, same_proc, 0, "gfx950_multi");
     descs[i] = {(const uint8_t*)src, reqs[i].dst_ptr,
                 (uint32_t)reqs[i].size};
   }

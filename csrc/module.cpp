@@ -195,6 +195,8 @@ PYBIND11_MODULE(_core, m) {
              d["gpu_rx"] = st.gpu_rx.load();
              d["cma_rx"] = st.cma_rx.load();
              d["unexpected_rx"] = st.unexpected_rx.load();
+             d["unexp_staged_bytes"] = st.unexp_staged_bytes.load();
+             d["deferred_sends"] = st.deferred_sends.load();
              return d;
            })
       .def("list_clients",
@@ -267,6 +269,8 @@ PYBIND11_MODULE(_core, m) {
              d["gpu_rx"] = st.gpu_rx.load();
              d["cma_rx"] = st.cma_rx.load();
              d["unexpected_rx"] = st.unexpected_rx.load();
+             d["unexp_staged_bytes"] = st.unexp_staged_bytes.load();
+             d["deferred_sends"] = st.deferred_sends.load();
              return d;
            })
       .def("evaluate_perf", [](PyClient& c, uint64_t msg_size) {
@@ -276,6 +280,14 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("gpu_available", [] { return gpu::available(); });
   m.def("gpu_device_count", [] { return gpu::device_count(); });
+  // IPC hygiene: close every imported hipIpc mapping and drop the handle
+  // caches. Call after returning GPU memory to the driver (e.g.
+  // torch.cuda.empty_cache()) so a reused base address can never serve a
+  // stale handle. Also registered as an atexit hook by the Python facade.
+  m.def("ipc_invalidate", [] {
+    py::gil_scoped_release rel;
+    gpu::ipc_close_all();
+  });
   // Test/bench hook: run the gfx950 copy kernel dst<-src synchronously.
   m.def("_copy_device_sync",
         [](uintptr_t dst, uintptr_t src, size_t nbytes, int device) {

@@ -52,7 +52,7 @@ try:
     from ._core import Client as _Client
     from ._core import Context, ServerEndpoint
     from ._core import Server as _Server
-    from ._core import gpu_available, gpu_device_count
+    from ._core import gpu_available, gpu_device_count, ipc_invalidate
 except ImportError as exc:  # pragma: no cover - build guidance
     raise ImportError(
         "starway_amd._core is not built. Run `python build_ext.py` (or "
@@ -72,6 +72,15 @@ def check_sys_libs() -> Literal["system"] | Literal["wheel"]:
 
 
 _context = Context()
+
+# IPC hygiene: imported hipIpc mappings are cached for the process lifetime
+# (registration-cache pattern, safe under torch's caching allocator). Close
+# them at interpreter exit; call ipc_invalidate() yourself after returning
+# GPU memory to the driver mid-run (e.g. torch.cuda.empty_cache()).
+if gpu_available():
+    import atexit
+
+    atexit.register(ipc_invalidate)
 
 
 class _CudaShim:
@@ -378,4 +387,5 @@ __all__ = [
     "list_benchmark_scenarios",
     "gpu_available",
     "gpu_device_count",
+    "ipc_invalidate",
 ]
