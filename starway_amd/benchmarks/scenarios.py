@@ -1,7 +1,11 @@
-"""Benchmark scenarios — same four workloads and metric definitions as the
-reference (benchmark.md / scenarios.py there), extended with an optional
-``device`` knob: ``cpu`` uses numpy buffers (reference behavior), ``cuda``
-uses torch HIP device tensors moved zero-copy over xGMI.
+"""Benchmark scenarios for the MI355X-native transport.
+
+The four workloads, their default knobs, and their metric names are the
+measurement contract shared with the reference (its benchmark.md defines
+them); everything below — the class-based scenario objects, the sampling
+clock, the buffer factory — is this repo's own implementation, extended
+with a ``device`` knob: ``cpu`` runs on numpy buffers, ``cuda``/``cuda:N``
+on torch HIP tensors moved zero-copy over xGMI.
 """
 from __future__ import annotations
 
@@ -14,10 +18,12 @@ import numpy as np
 
 TAG_MASK: int = (1 << 64) - 1
 
+# Control-plane tags (bench.py session protocol).
 CONTROL_TAG = 0x1AA0
 READY_TAG = 0x1AA1
 DONE_TAG = 0x1AA2
 
+# Data-plane tags, one block per scenario.
 LARGE_DATA_TAG = 0x2B00
 SMALL_DATA_TAG = 0x2B10
 SMALL_ACK_TAG = 0x2B11
@@ -75,15 +81,14 @@ class ScenarioDefinition:
     server_runner: ServerRunner
 
 
-def _merged(defaults: Mapping[str, Any], overrides: Mapping[str, Any]) -> Dict[str, Any]:
-    out = dict(defaults)
-    out.update({k: v for k, v in overrides.items() if v is not None})
-    return out
+# ---------------------------------------------------------------------------
+# shared machinery
+# ---------------------------------------------------------------------------
 
-
-def _alloc(nbytes: int, device: str, fill: int | None = None):
-    """Message buffer on the requested device ('cpu' -> numpy uint8,
-    'cuda' / 'cuda:N' -> torch uint8 HIP tensor)."""
+def buffer_on(device: str, nbytes: int, fill: int | None = None):
+    """Allocate a message buffer: numpy uint8 on 'cpu', torch uint8 HIP
+    tensor on 'cuda'/'cuda:N' (synchronized so timing never includes the
+    fill kernel)."""
     if device == "cpu":
         arr = np.empty(nbytes, dtype=np.uint8)
         if fill is not None:
@@ -98,290 +103,309 @@ def _alloc(nbytes: int, device: str, fill: int | None = None):
     return t
 
 
+class SampleClock:
+    """Wall-clock sampler: lap() times one unit of work, keeping samples
+    only past the warmup count; derived stats come out of laps()."""
+
+    def __init__(self, warmup: int) -> None:
+        self._warmup = warmup
+        self._seen = 0
+        self._t0 = 0.0
+        self.samples: List[float] = []
+
+    def __enter__(self) -> "SampleClock":
+        self._t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc) -> None:
+        dt = time.perf_counter() - self._t0
+        self._seen += 1
+        if self._seen > self._warmup:
+            self.samples.append(dt)
+
+    @property
+    def total(self) -> float:
+        return sum(self.samples)
+
+    def us(self) -> np.ndarray:
+        return (np.asarray(self.samples) if self.samples else np.zeros(1)) * 1e6
+
+
+def _rate_gbps(nbytes: float, seconds: float) -> float:
+    return (nbytes / seconds) / 1e9 if seconds > 0 else 0.0
+
+
+class Scenario:
+    """One benchmark workload: subclasses set the class attributes and
+    implement drive() (client side, returns the result) and serve()
+    (server side). resolve() folds CLI overrides onto the defaults."""
+
+    name: str = ""
+    describe: str = ""
+    knobs: Dict[str, Any] = {}
+
+    @classmethod
+    def resolve(cls, overrides: Mapping[str, Any]) -> Dict[str, Any]:
+        cfg = dict(cls.knobs)
+        for key, val in overrides.items():
+            if val is not None:
+                cfg[key] = val
+        return cfg
+
+    @classmethod
+    def repetitions(cls, cfg: Mapping[str, Any]) -> tuple[int, int]:
+        warm = int(cfg.get("warmup", cfg.get("warmup_batches", 0)))
+        return warm, int(cfg["iterations"])
+
+    async def drive(self, ctx: ClientRuntime, cfg: Mapping[str, Any]) -> ScenarioResult:
+        raise NotImplementedError
+
+    async def serve(self, ctx: ServerRuntime, cfg: Mapping[str, Any]) -> None:
+        raise NotImplementedError
+
+
 # ---------------------------------------------------------------------------
-# large-array: one-way bandwidth of a single big message
+# large-array: one-way bandwidth of a single big message (flush-inclusive)
 # ---------------------------------------------------------------------------
 
-async def _large_array_client(ctx: ClientRuntime, config: Mapping[str, Any]) -> ScenarioResult:
-    cfg = _merged(LARGE_ARRAY.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    warmup = int(cfg["warmup"])
-    iterations = int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
+class LargeArrayScenario(Scenario):
+    name = "large-array"
+    describe = "Measure one-way bandwidth by transferring a single large buffer."
+    knobs = {"message_bytes": 1 << 30, "warmup": 1, "iterations": 3,
+             "device": "cpu"}
 
-    payload = _alloc(message_bytes, device, fill=0x5A)
-    durations: list[float] = []
-    per_iter_gbps: list[float] = []
+    async def drive(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        warm, iters = self.repetitions(cfg)
+        payload = buffer_on(str(cfg["device"]), nbytes, fill=0x5A)
 
-    for idx in range(warmup + iterations):
-        start = time.perf_counter()
-        await ctx.client.asend(payload, LARGE_DATA_TAG)
-        await ctx.flush()
-        elapsed = time.perf_counter() - start
-        if idx >= warmup:
-            durations.append(elapsed)
-            if elapsed > 0:
-                per_iter_gbps.append((message_bytes / elapsed) / 1e9)
+        clock = SampleClock(warm)
+        for _ in range(warm + iters):
+            with clock:
+                await ctx.client.asend(payload, LARGE_DATA_TAG)
+                await ctx.flush()
 
-    total_time = sum(durations)
-    metrics = {
-        "total_seconds": total_time,
-        "avg_seconds_per_iter": total_time / iterations if iterations else 0.0,
-        "avg_gbps": (message_bytes * iterations / total_time) / 1e9 if total_time else 0.0,
-        "best_gbps": max(per_iter_gbps, default=0.0),
-        "worst_gbps": min(per_iter_gbps, default=0.0),
-    }
-    return ScenarioResult(
-        name="large-array",
-        metrics=metrics,
-        samples={"duration_seconds": durations, "per_iter_gbps": per_iter_gbps},
-        config=dict(cfg),
-    )
-
-
-async def _large_array_server(ctx: ServerRuntime, config: Mapping[str, Any]) -> None:
-    cfg = _merged(LARGE_ARRAY.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    total = int(cfg["warmup"]) + int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
-    recv_buffer = _alloc(message_bytes, device)
-    await ctx.signal_ready()
-    for _ in range(total):
-        await ctx.server.arecv(recv_buffer, LARGE_DATA_TAG, ctx.tag_mask)
-    await ctx.flush_endpoint()
-
-
-# ---------------------------------------------------------------------------
-# small-messages: many concurrent small sends
-# ---------------------------------------------------------------------------
-
-async def _small_messages_client(ctx: ClientRuntime, config: Mapping[str, Any]) -> ScenarioResult:
-    cfg = _merged(SMALL_MESSAGES.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    warmup = int(cfg["warmup_batches"])
-    iterations = int(cfg["iterations"])
-    concurrency = int(cfg["concurrency"])
-    device = str(cfg.get("device", "cpu"))
-
-    payloads = [_alloc(message_bytes, device, fill=i % 251) for i in range(concurrency)]
-    durations: list[float] = []
-    per_message_latency: list[float] = []
-
-    for batch in range(warmup + iterations):
-        start = time.perf_counter()
-        await asyncio.gather(
-            *(ctx.client.asend(buf, SMALL_DATA_TAG) for buf in payloads)
+        rates = [_rate_gbps(nbytes, dt) for dt in clock.samples]
+        return ScenarioResult(
+            name=self.name,
+            metrics={
+                "total_seconds": clock.total,
+                "avg_seconds_per_iter": clock.total / iters if iters else 0.0,
+                "avg_gbps": _rate_gbps(nbytes * iters, clock.total),
+                "best_gbps": max(rates, default=0.0),
+                "worst_gbps": min(rates, default=0.0),
+            },
+            samples={"duration_seconds": clock.samples,
+                     "per_iter_gbps": rates},
+            config=cfg,
         )
-        await ctx.flush()
-        elapsed = time.perf_counter() - start
-        if batch >= warmup:
-            durations.append(elapsed)
-            if concurrency:
-                per_message_latency.append(elapsed / concurrency)
 
-    total_messages = iterations * concurrency
-    total_time = sum(durations)
-    lat_us = np.array(per_message_latency) * 1e6 if per_message_latency else np.zeros(1)
-    metrics = {
-        "total_seconds": total_time,
-        "messages_per_second": total_messages / total_time if total_time else 0.0,
-        "bandwidth_gbps": (message_bytes * total_messages / total_time) / 1e9
-        if total_time
-        else 0.0,
-        "latency_p50_us": float(np.percentile(lat_us, 50)),
-        "latency_p95_us": float(np.percentile(lat_us, 95)),
-    }
-    return ScenarioResult(
-        name="small-messages",
-        metrics=metrics,
-        samples={
-            "batch_duration_seconds": durations,
-            "avg_latency_seconds": per_message_latency,
-        },
-        config=dict(cfg),
-    )
+    async def serve(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        warm, iters = self.repetitions(cfg)
+        sink = buffer_on(str(cfg["device"]), int(cfg["message_bytes"]))
+        await ctx.signal_ready()
+        for _ in range(warm + iters):
+            await ctx.server.arecv(sink, LARGE_DATA_TAG, ctx.tag_mask)
+        await ctx.flush_endpoint()
 
 
-async def _small_messages_server(ctx: ServerRuntime, config: Mapping[str, Any]) -> None:
-    cfg = _merged(SMALL_MESSAGES.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    total = int(cfg["warmup_batches"]) + int(cfg["iterations"])
-    concurrency = int(cfg["concurrency"])
-    device = str(cfg.get("device", "cpu"))
-    buffers = [_alloc(message_bytes, device) for _ in range(concurrency)]
-    await ctx.signal_ready()
-    for _ in range(total):
-        await asyncio.gather(
-            *(ctx.server.arecv(buf, SMALL_DATA_TAG, ctx.tag_mask) for buf in buffers)
+# ---------------------------------------------------------------------------
+# small-messages: batches of concurrent small sends, rate + latency
+# ---------------------------------------------------------------------------
+
+class SmallMessagesScenario(Scenario):
+    name = "small-messages"
+    describe = "Stress many small messages with configurable concurrency."
+    knobs = {"message_bytes": 1024, "warmup_batches": 2, "iterations": 10,
+             "concurrency": 64, "device": "cpu"}
+
+    async def drive(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        fanout = int(cfg["concurrency"])
+        warm, iters = self.repetitions(cfg)
+        wave = [buffer_on(str(cfg["device"]), nbytes, fill=i % 251)
+                for i in range(fanout)]
+
+        clock = SampleClock(warm)
+        for _ in range(warm + iters):
+            with clock:
+                await asyncio.gather(
+                    *(ctx.client.asend(buf, SMALL_DATA_TAG) for buf in wave))
+                await ctx.flush()
+
+        # Batch-time / concurrency approximation of per-message latency
+        # (the reference's small-messages metric definition).
+        per_msg = [dt / fanout for dt in clock.samples] if fanout else []
+        lat_us = (np.asarray(per_msg) if per_msg else np.zeros(1)) * 1e6
+        sent = iters * fanout
+        return ScenarioResult(
+            name=self.name,
+            metrics={
+                "total_seconds": clock.total,
+                "messages_per_second": sent / clock.total if clock.total else 0.0,
+                "bandwidth_gbps": _rate_gbps(nbytes * sent, clock.total),
+                "latency_p50_us": float(np.percentile(lat_us, 50)),
+                "latency_p95_us": float(np.percentile(lat_us, 95)),
+            },
+            samples={"batch_duration_seconds": clock.samples,
+                     "avg_latency_seconds": per_msg},
+            config=cfg,
         )
-    await ctx.flush_endpoint()
+
+    async def serve(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        fanout = int(cfg["concurrency"])
+        warm, iters = self.repetitions(cfg)
+        sinks = [buffer_on(str(cfg["device"]), int(cfg["message_bytes"]))
+                 for _ in range(fanout)]
+        await ctx.signal_ready()
+        for _ in range(warm + iters):
+            await asyncio.gather(
+                *(ctx.server.arecv(b, SMALL_DATA_TAG, ctx.tag_mask)
+                  for b in sinks))
+        await ctx.flush_endpoint()
 
 
 # ---------------------------------------------------------------------------
-# pingpong-flag: 1-byte round-trip latency
+# pingpong-flag: tiny-message round-trip latency (pre-posted recv)
 # ---------------------------------------------------------------------------
 
-async def _pingpong_client(ctx: ClientRuntime, config: Mapping[str, Any]) -> ScenarioResult:
-    cfg = _merged(PINGPONG_FLAG.defaults, config)
-    warmup = int(cfg["warmup"])
-    iterations = int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
-    message_bytes = int(cfg.get("message_bytes", 1))
+class PingpongFlagScenario(Scenario):
+    name = "pingpong-flag"
+    describe = "Round-trip a single-byte control flag to capture latency."
+    knobs = {"warmup": 100, "iterations": 1000, "message_bytes": 1,
+             "device": "cpu"}
 
-    send_buf = _alloc(message_bytes, device, fill=1)
-    recv_buf = _alloc(message_bytes, device, fill=0)
-    durations: list[float] = []
+    async def drive(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        warm, iters = self.repetitions(cfg)
+        ping = buffer_on(str(cfg["device"]), nbytes, fill=1)
+        pong = buffer_on(str(cfg["device"]), nbytes, fill=0)
 
-    for _ in range(warmup):
-        fut = ctx.client.arecv(recv_buf, FLAG_PONG_TAG, ctx.tag_mask)
-        await ctx.client.asend(send_buf, FLAG_PING_TAG)
-        await fut
+        clock = SampleClock(warm)
+        for _ in range(warm + iters):
+            # Post the reply recv before sending so the pong never lands
+            # unexpected — this is the latency-path contract.
+            reply = ctx.client.arecv(pong, FLAG_PONG_TAG, ctx.tag_mask)
+            with clock:
+                await ctx.client.asend(ping, FLAG_PING_TAG)
+                await reply
+        await ctx.flush()
 
-    for _ in range(iterations):
-        fut = ctx.client.arecv(recv_buf, FLAG_PONG_TAG, ctx.tag_mask)
-        start = time.perf_counter()
-        await ctx.client.asend(send_buf, FLAG_PING_TAG)
-        await fut
-        durations.append(time.perf_counter() - start)
+        rtt = clock.us()
+        return ScenarioResult(
+            name=self.name,
+            metrics={
+                "avg_rtt_us": float(rtt.mean()),
+                "median_rtt_us": float(np.median(rtt)),
+                "min_rtt_us": float(rtt.min()),
+                "max_rtt_us": float(rtt.max()),
+                "avg_one_way_us": float(rtt.mean()) / 2.0,
+            },
+            samples={"rtt_seconds": clock.samples},
+            config=cfg,
+        )
 
-    await ctx.flush()
-    lat = np.array(durations) * 1e6 if durations else np.zeros(1)
-    metrics = {
-        "avg_rtt_us": float(np.mean(lat)),
-        "median_rtt_us": float(np.median(lat)),
-        "min_rtt_us": float(np.min(lat)),
-        "max_rtt_us": float(np.max(lat)),
-        "avg_one_way_us": float(np.mean(lat)) / 2.0,
-    }
-    return ScenarioResult(
-        name="pingpong-flag",
-        metrics=metrics,
-        samples={"rtt_seconds": durations},
-        config=dict(cfg),
+    async def serve(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        warm, iters = self.repetitions(cfg)
+        flag = buffer_on(str(cfg["device"]), nbytes, fill=0)
+        ack = buffer_on(str(cfg["device"]), nbytes, fill=1)
+        await ctx.signal_ready()
+        for _ in range(warm + iters):
+            await ctx.server.arecv(flag, FLAG_PING_TAG, ctx.tag_mask)
+            await ctx.server.asend(ctx.endpoint, ack, FLAG_PONG_TAG)
+        await ctx.flush_endpoint()
+
+
+# ---------------------------------------------------------------------------
+# streaming-duplex: both directions stream concurrently
+# ---------------------------------------------------------------------------
+
+class StreamingDuplexScenario(Scenario):
+    name = "streaming-duplex"
+    describe = "Bidirectional medium-sized streaming in both directions."
+    knobs = {"message_bytes": 4 * 1024 * 1024, "warmup": 8, "iterations": 64,
+             "device": "cpu"}
+
+    async def drive(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        warm, iters = self.repetitions(cfg)
+        up = buffer_on(str(cfg["device"]), nbytes, fill=0x7B)
+        down = buffer_on(str(cfg["device"]), nbytes)
+
+        clock = SampleClock(warm)
+        for _ in range(warm + iters):
+            inbound = ctx.client.arecv(down, STREAM_DOWN_TAG, ctx.tag_mask)
+            with clock:
+                await asyncio.gather(
+                    ctx.client.asend(up, STREAM_UP_TAG), inbound)
+        await ctx.flush()
+
+        one_way = nbytes * iters
+        return ScenarioResult(
+            name=self.name,
+            metrics={
+                "total_seconds": clock.total,
+                "avg_seconds_per_iter": clock.total / iters if iters else 0.0,
+                "client_to_server_gbps": _rate_gbps(one_way, clock.total),
+                "server_to_client_gbps": _rate_gbps(one_way, clock.total),
+                "aggregate_gbps": _rate_gbps(2 * one_way, clock.total),
+            },
+            samples={"iteration_seconds": clock.samples},
+            config=cfg,
+        )
+
+    async def serve(self, ctx, overrides):
+        cfg = self.resolve(overrides)
+        nbytes = int(cfg["message_bytes"])
+        warm, iters = self.repetitions(cfg)
+        down = buffer_on(str(cfg["device"]), nbytes, fill=0x3C)
+        up = buffer_on(str(cfg["device"]), nbytes)
+        await ctx.signal_ready()
+        for _ in range(warm + iters):
+            await asyncio.gather(
+                ctx.server.arecv(up, STREAM_UP_TAG, ctx.tag_mask),
+                ctx.server.asend(ctx.endpoint, down, STREAM_DOWN_TAG))
+        await ctx.flush_endpoint()
+
+
+# ---------------------------------------------------------------------------
+# registry
+# ---------------------------------------------------------------------------
+
+def _definition(cls: type[Scenario]) -> ScenarioDefinition:
+    inst = cls()
+    return ScenarioDefinition(
+        name=cls.name,
+        description=cls.describe,
+        defaults=dict(cls.knobs),
+        client_runner=inst.drive,
+        server_runner=inst.serve,
     )
 
-
-async def _pingpong_server(ctx: ServerRuntime, config: Mapping[str, Any]) -> None:
-    cfg = _merged(PINGPONG_FLAG.defaults, config)
-    total = int(cfg["warmup"]) + int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
-    message_bytes = int(cfg.get("message_bytes", 1))
-    recv_buf = _alloc(message_bytes, device, fill=0)
-    ack_buf = _alloc(message_bytes, device, fill=1)
-    await ctx.signal_ready()
-    for _ in range(total):
-        await ctx.server.arecv(recv_buf, FLAG_PING_TAG, ctx.tag_mask)
-        await ctx.server.asend(ctx.endpoint, ack_buf, FLAG_PONG_TAG)
-    await ctx.flush_endpoint()
-
-
-# ---------------------------------------------------------------------------
-# streaming-duplex: concurrent streams both directions
-# ---------------------------------------------------------------------------
-
-async def _streaming_duplex_client(ctx: ClientRuntime, config: Mapping[str, Any]) -> ScenarioResult:
-    cfg = _merged(STREAMING_DUPLEX.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    warmup = int(cfg["warmup"])
-    iterations = int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
-
-    send_buf = _alloc(message_bytes, device, fill=0x7B)
-    recv_buf = _alloc(message_bytes, device)
-    durations: list[float] = []
-
-    for idx in range(warmup + iterations):
-        recv_future = ctx.client.arecv(recv_buf, STREAM_DOWN_TAG, ctx.tag_mask)
-        start = time.perf_counter()
-        send_future = ctx.client.asend(send_buf, STREAM_UP_TAG)
-        await asyncio.gather(send_future, recv_future)
-        elapsed = time.perf_counter() - start
-        if idx >= warmup:
-            durations.append(elapsed)
-
-    await ctx.flush()
-    total_time = sum(durations)
-    per_dir = message_bytes * iterations
-    metrics = {
-        "total_seconds": total_time,
-        "avg_seconds_per_iter": total_time / iterations if iterations else 0.0,
-        "client_to_server_gbps": per_dir / total_time / 1e9 if total_time else 0.0,
-        "server_to_client_gbps": per_dir / total_time / 1e9 if total_time else 0.0,
-        "aggregate_gbps": 2 * per_dir / total_time / 1e9 if total_time else 0.0,
-    }
-    return ScenarioResult(
-        name="streaming-duplex",
-        metrics=metrics,
-        samples={"iteration_seconds": durations},
-        config=dict(cfg),
-    )
-
-
-async def _streaming_duplex_server(ctx: ServerRuntime, config: Mapping[str, Any]) -> None:
-    cfg = _merged(STREAMING_DUPLEX.defaults, config)
-    message_bytes = int(cfg["message_bytes"])
-    total = int(cfg["warmup"]) + int(cfg["iterations"])
-    device = str(cfg.get("device", "cpu"))
-    send_buf = _alloc(message_bytes, device, fill=0x3C)
-    recv_buf = _alloc(message_bytes, device)
-    await ctx.signal_ready()
-    for _ in range(total):
-        recv_future = ctx.server.arecv(recv_buf, STREAM_UP_TAG, ctx.tag_mask)
-        send_future = ctx.server.asend(ctx.endpoint, send_buf, STREAM_DOWN_TAG)
-        await asyncio.gather(recv_future, send_future)
-    await ctx.flush_endpoint()
-
-
-# ---------------------------------------------------------------------------
-
-LARGE_ARRAY = ScenarioDefinition(
-    name="large-array",
-    description="Measure one-way bandwidth by transferring a single large buffer.",
-    defaults={"message_bytes": 1 << 30, "warmup": 1, "iterations": 3, "device": "cpu"},
-    client_runner=_large_array_client,
-    server_runner=_large_array_server,
-)
-
-SMALL_MESSAGES = ScenarioDefinition(
-    name="small-messages",
-    description="Stress many small messages with configurable concurrency.",
-    defaults={
-        "message_bytes": 1024,
-        "warmup_batches": 2,
-        "iterations": 10,
-        "concurrency": 64,
-        "device": "cpu",
-    },
-    client_runner=_small_messages_client,
-    server_runner=_small_messages_server,
-)
-
-PINGPONG_FLAG = ScenarioDefinition(
-    name="pingpong-flag",
-    description="Round-trip a single-byte control flag to capture latency.",
-    defaults={"warmup": 100, "iterations": 1000, "message_bytes": 1, "device": "cpu"},
-    client_runner=_pingpong_client,
-    server_runner=_pingpong_server,
-)
-
-STREAMING_DUPLEX = ScenarioDefinition(
-    name="streaming-duplex",
-    description="Bidirectional medium-sized streaming in both directions.",
-    defaults={"message_bytes": 4 * 1024 * 1024, "warmup": 8, "iterations": 64, "device": "cpu"},
-    client_runner=_streaming_duplex_client,
-    server_runner=_streaming_duplex_server,
-)
 
 SCENARIOS: Dict[str, ScenarioDefinition] = {
-    s.name: s for s in (LARGE_ARRAY, SMALL_MESSAGES, PINGPONG_FLAG, STREAMING_DUPLEX)
+    cls.name: _definition(cls)
+    for cls in (LargeArrayScenario, SmallMessagesScenario,
+                PingpongFlagScenario, StreamingDuplexScenario)
 }
 
 __all__ = [
     "SCENARIOS",
+    "Scenario",
     "ScenarioDefinition",
     "ScenarioResult",
     "ClientRuntime",
     "ServerRuntime",
+    "SampleClock",
+    "buffer_on",
     "CONTROL_TAG",
     "READY_TAG",
     "DONE_TAG",
