@@ -17,7 +17,7 @@ CSRC = REPO / "csrc"
 BUILD = REPO / "build"
 PKG = REPO / "starway_amd"
 
-SOURCES = ["engine.cpp", "gpu.cpp", "module.cpp", "kernels.hip"]
+SOURCES = ["engine.cpp", "gpu.cpp", "module.cpp", "kernels.hip", "smallmsg.hip"]
 RCCL_SOURCES = ["rccl_group.cpp"]
 
 HIPCC = os.environ.get("STARWAY_HIPCC", "hipcc")

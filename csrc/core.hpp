@@ -53,6 +53,11 @@ enum FrameType : uint16_t {
   FT_SHM_ACK = 11,     // flags: 0 accept / 1 decline; sender's LAST tcp frame
   FT_SHM_SWITCH = 12,  // server's LAST tcp frame after ACK
   FT_RTS_CPU = 13,     // payload: CmaDesc — large same-host CPU rendezvous
+  // Small-message inbox plane (smallmsg.hip): sender pushes payloads into
+  // the peer's device ring over xGMI; these frames carry only control.
+  FT_INBOX_OFFER = 14,   // payload: gpu::InboxInfo of MY ring (peer pushes)
+  FT_INBOX_CREDIT = 15,  // hdr.aux = consumed seq (slots <= aux are free)
+  FT_SMSG = 16,          // hdr.tag/aux/op_id = tag / size / ring seq
 };
 
 #pragma pack(push, 1)
@@ -69,6 +74,21 @@ struct FrameHeader {
 static_assert(sizeof(FrameHeader) == 40);
 
 constexpr int kIpcHandleBytes = 64;  // HIP_IPC_HANDLE_SIZE
+
+namespace gpu {
+// Small-message inbox descriptor (smallmsg.hip): a device ring the OWNER
+// allocates and exports; the peer pushes payloads into it over xGMI. Sent
+// raw as the FT_INBOX_OFFER payload (same-host peers share the ABI).
+constexpr uint32_t kInboxHdrBytes = 64;  // [seq][tag][size][pad] per slot
+
+struct InboxInfo {
+  uint64_t base = 0;   // device pointer in the OWNER's address space
+  int32_t device = -1; // owner's device ordinal
+  uint32_t slots = 0;
+  uint32_t slot_bytes = 0;  // header + padded payload capacity
+  uint8_t handle[kIpcHandleBytes] = {0};
+};
+}  // namespace gpu
 
 // GPU rendezvous descriptor — carried as RTS payload. The receiver pulls the
 // data over xGMI (peer kernel copy) or locally, then acks with RECV_DONE.
@@ -264,6 +284,8 @@ struct UnexpectedMsg {
   Connection* conn = nullptr;
   bool is_rts = false;
   bool is_cma = false;
+  bool is_smsg = false;     // payload parked in our inbox slot
+  uint64_t smsg_seq = 0;
   uint64_t sender_op_id = 0;
   RtsDesc rts{};
   CmaDesc cma{};
@@ -324,6 +346,20 @@ struct Connection {
   // start as acks arrive (in-flight window sized for 288 GB HBM3E).
   uint64_t inflight_rndv_bytes = 0;
   std::deque<Op*> deferred_sends;
+
+  // --- small-message inbox (smallmsg.hip) ---
+  // Local ring (peer pushes into it):
+  gpu::InboxInfo inbox_l;
+  bool inbox_l_active = false;
+  uint64_t inbox_seen_seq = 0;      // last SMSG sequence processed
+  uint64_t inbox_consumed = 0;      // contiguous consumed prefix
+  uint64_t inbox_credited = 0;      // last consumed value sent as CREDIT
+  std::set<uint64_t> inbox_released;  // out-of-order consumed seqs
+  // Remote ring (we push into it):
+  gpu::InboxInfo inbox_r;
+  bool inbox_r_active = false;
+  uint64_t inbox_next_seq = 1;      // next sequence we will assign
+  uint64_t inbox_credit_base = 0;   // peer-consumed seq (slot availability)
 
   // --- tx ---
   std::deque<TxItem> txq;
@@ -395,6 +431,8 @@ class Engine {
     std::atomic<uint64_t> msgs_sent{0}, msgs_received{0};
     std::atomic<uint64_t> bytes_sent{0}, bytes_received{0};
     std::atomic<uint64_t> eager_rx{0}, gpu_rx{0}, cma_rx{0};
+    std::atomic<uint64_t> inbox_rx{0}, inbox_tx{0};  // small-message plane
+    std::atomic<uint64_t> doorbell_rx{0};            // of inbox_rx, pre-armed
     std::atomic<uint64_t> unexpected_rx{0};
     std::atomic<uint64_t> unexp_staged_bytes{0};  // flow-control watermark
     std::atomic<uint64_t> deferred_sends{0};      // window-queued sends
@@ -521,6 +559,65 @@ class Engine {
   };
   std::vector<std::unique_ptr<CmaPull>> cma_pulls_;
 
+  // --- small-message inbox machinery (engine thread only) ---
+  struct PendingPush {
+    Op* op;
+    uint64_t seq;
+  };
+  std::map<Connection*, std::vector<PendingPush>> pending_pushes_;
+  struct PushBatch {
+    void* ticket = nullptr;
+    Connection* conn = nullptr;
+    std::vector<Op*> ops;
+  };
+  std::vector<std::unique_ptr<PushBatch>> push_batches_;
+  struct PendingUnpack {
+    Connection* conn;
+    uint64_t seq;
+    uint64_t size;
+    uint64_t tag;
+    Op* recv_op;
+    uint8_t* dst;  // device dst, or null => pinned bounce
+  };
+  std::vector<PendingUnpack> pending_unpacks_;
+  struct UnpackBatch {
+    void* ticket = nullptr;
+    Connection* conn = nullptr;
+    std::vector<PendingUnpack> msgs;
+    std::vector<bool> done;
+    size_t remaining = 0;
+  };
+  std::vector<std::unique_ptr<UnpackBatch>> unpack_batches_;
+  // Doorbell: at most one pre-armed wait kernel per engine.
+  struct Armed {
+    void* ticket = nullptr;
+    Op* recv_op = nullptr;
+    Connection* conn = nullptr;
+    uint64_t seq = 0;
+  } armed_;
+  // A doorbell copy whose SMSG control frame has not arrived yet (the
+  // kernel consumed the message during a disarm race).
+  struct ArmedDone {
+    bool active = false;
+    Connection* conn = nullptr;
+    uint64_t seq = 0;
+    uint64_t size = 0;
+    Op* recv_op = nullptr;
+  } armed_done_;
+  void handle_smsg(Connection* c, uint64_t tag, uint64_t size, uint64_t seq);
+  void dispatch_smsg_to_recv(Connection* c, Op* r, uint64_t tag,
+                             uint64_t size, uint64_t seq);
+  void flush_pending_pushes();
+  void progress_pushes(bool& did_work);
+  void flush_pending_unpacks();
+  void progress_unpacks(bool& did_work);
+  void progress_armed(bool& did_work);
+  void inbox_release_seq(Connection* c, uint64_t seq);
+  void try_arm();
+  bool steal_armed(Op* r);  // false => the doorbell consumed r
+  Op* take_matching_recv(uint64_t tag);
+  void repost_recv_front(Op* r);  // disarms, then push_front
+
   // Connect state (client).
   std::string connect_host_;
   int connect_port_ = 0;
@@ -582,6 +679,49 @@ void attach_bounce(void* ticket, RawBuf&& bounce);
 // caches (process-exit hook + explicit invalidation for allocators that
 // return memory to the driver, e.g. torch empty_cache).
 void ipc_close_all();
+
+// ---------------------------------------------------------------------------
+// Small-message inbox data plane (smallmsg.hip): sender-push ring in the
+// receiver's HBM, written over xGMI; local unpack or pre-armed doorbell
+// kernel on the receiver. See smallmsg.hip header comment for the design;
+// InboxInfo / kInboxHdrBytes are declared above (wire ABI).
+// ---------------------------------------------------------------------------
+bool inbox_create(InboxInfo* out, std::string* err);
+void inbox_destroy(const InboxInfo& ib);
+
+struct PushMsg {
+  const uint8_t* src;
+  uint32_t size;
+  uint64_t seq;
+  uint64_t tag;
+};
+// Batched push into the PEER's inbox; returns an event ticket polled with
+// push_poll / freed with push_free.
+void* inbox_push(const InboxInfo& peer, bool same_proc, int run_device,
+                 const PushMsg* msgs, int n, std::string* err);
+int push_poll(void* ticket, std::string* err);
+void push_free(void* ticket);
+
+struct UnpackMsg {
+  uint64_t seq;
+  uint32_t size;
+  uint8_t* dst;  // device pointer, or null => pinned host bounce
+};
+void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
+                   std::string* err);
+int unpack_poll(void* ticket, int idx, std::string* err);  // 0/1/-1
+const uint8_t* unpack_bounce(void* ticket, int idx);
+void unpack_free(void* ticket);
+
+// Doorbell: pre-armed bounded wait for the next sequence number, copying
+// into dst on a tag match. arm_poll: 0 running / 1 copied / 2 nomatch /
+// 3 canceled / 4 expired.
+void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
+               uint64_t mask, uint8_t* dst, uint64_t max_size,
+               std::string* err);
+int arm_poll(void* ticket, uint64_t* size_out);
+void arm_cancel(void* ticket);
+void arm_free(void* ticket);
 // Poll a ticket: 1 done, 0 pending, -1 error.
 int poll_ticket(void* ticket, std::string* err);
 void free_ticket(void* ticket);

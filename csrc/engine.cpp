@@ -721,8 +721,10 @@ void Engine::loop_iteration(bool& did_work) {
       on_conn_dead(c.get());
   }
   int timeout = 0;
-  bool busy =
-      !gpu_pulls_.empty() || !cma_pulls_.empty() || !d2h_sends_.empty();
+  bool busy = !gpu_pulls_.empty() || !cma_pulls_.empty() ||
+              !d2h_sends_.empty() || !push_batches_.empty() ||
+              !unpack_batches_.empty() || !pending_pushes_.empty() ||
+              !pending_unpacks_.empty() || armed_done_.active;
   if (!busy)
     for (auto& c : conns_)
       if (c->want_write() ||
@@ -748,6 +750,18 @@ void Engine::loop_iteration(bool& did_work) {
     flush_small_pulls();
     did_work = true;
   }
+  if (!pending_pushes_.empty()) {
+    flush_pending_pushes();
+    did_work = true;
+  }
+  if (!pending_unpacks_.empty()) {
+    flush_pending_unpacks();
+    did_work = true;
+  }
+  if (!push_batches_.empty()) progress_pushes(did_work);
+  if (!unpack_batches_.empty()) progress_unpacks(did_work);
+  if (armed_.ticket) progress_armed(did_work);
+  try_arm();
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
   if (!cma_pulls_.empty()) progress_cma(did_work);
   if (!d2h_sends_.empty()) progress_d2h(did_work);
@@ -941,6 +955,24 @@ void Engine::start_send(Op* op, Connection* c) {
   }
   static const bool force_xhost =
       getenv("STARWAY_FORCE_XHOST") != nullptr;  // test hook
+  // Small-message inbox push: device payloads up to the slot capacity are
+  // written straight into the peer's ring over xGMI by a batched push
+  // kernel — no RTS round trip, no per-message event, no RECV_DONE. The
+  // FT_SMSG control frame is enqueued HERE so per-connection message order
+  // is preserved against eager/RTS traffic. The op completes when the push
+  // kernel's event lands (payload captured out of the user buffer).
+  if (op->buf.device >= 0 && op->buf.rows == 0 && op->buf.size > 0 &&
+      !force_xhost && c->inbox_r_active &&
+      op->buf.size <= c->inbox_r.slot_bytes - gpu::kInboxHdrBytes &&
+      c->inbox_next_seq <= c->inbox_credit_base + c->inbox_r.slots) {
+    uint64_t seq = c->inbox_next_seq++;
+    enqueue_frame(c, FT_SMSG, op->tag, seq, op->buf.size, nullptr, 0, false);
+    op->owned_by_d2h = true;  // reaped by the push machinery, not conn-death
+    gpu_sends_[op->id] = op;  // flush coverage until the push ticket lands
+    pending_pushes_[c].push_back(PendingPush{op, seq});
+    stats_.inbox_tx.fetch_add(1, std::memory_order_relaxed);
+    return;
+  }
   if (op->buf.device >= 0 &&
       (force_xhost || memcmp(c->peer.host_id, host_id(), 16) != 0)) {
     // Cross-host GPU send: hipIpc cannot cross hosts — stage the payload
@@ -1196,11 +1228,18 @@ void Engine::on_frame(Connection* c) {
     case FT_EAGER:
       begin_eager(c);
       return;
+    case FT_SMSG:
+      handle_smsg(c, h.tag, h.aux, h.op_id);
+      return;
+    case FT_INBOX_CREDIT:
+      if (h.aux > c->inbox_credit_base) c->inbox_credit_base = h.aux;
+      return;
     case FT_HELLO:
     case FT_RTS:
     case FT_RTS_CPU:
     case FT_RECV_FAIL:
     case FT_SHM_OFFER:
+    case FT_INBOX_OFFER:
       if (h.size > (16 << 20)) {
         on_conn_dead(c);
         return;
@@ -1287,14 +1326,9 @@ void Engine::on_frame_payload(Connection* c) {
       // Match like any other message; unmatched waits in the unexpected
       // queue as a descriptor.
       bool matched = false;
-      for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
-        Op* r = *it;
-        if ((h.tag & r->tag_mask) == (r->tag & r->tag_mask)) {
-          posted_recvs_.erase(it);
-          start_cma_pull(r, cma, h.tag, h.aux, h.op_id, c);
-          matched = true;
-          break;
-        }
+      if (Op* r = take_matching_recv(h.tag)) {
+        start_cma_pull(r, cma, h.tag, h.aux, h.op_id, c);
+        matched = true;
       }
       if (!matched) {
         auto um = std::make_unique<UnexpectedMsg>();
@@ -1312,6 +1346,18 @@ void Engine::on_frame_payload(Connection* c) {
     case FT_RECV_FAIL: {
       std::string err((const char*)c->rx_small.data(), c->rx_small.size());
       on_gpu_send_acked(h.op_id, true, err);
+      break;
+    }
+    case FT_INBOX_OFFER: {
+      if (c->rx_small.size() != sizeof(gpu::InboxInfo)) break;
+      memcpy(&c->inbox_r, c->rx_small.data(), sizeof(gpu::InboxInfo));
+      if (c->inbox_r.slots && c->inbox_r.slot_bytes > gpu::kInboxHdrBytes &&
+          gpu::available()) {
+        c->inbox_r_active = true;
+        c->inbox_next_seq = 1;
+        c->inbox_credit_base = 0;
+        if (c->ep) c->ep->transports.emplace_back("xgmi", "inbox_push");
+      }
       break;
     }
     case FT_SHM_OFFER: {
@@ -1417,6 +1463,24 @@ void Engine::on_hello(Connection* c) {
       complete(std::move(comp));
     }
   }
+  // Small-message inbox: both sides offer their ring to any same-host GPU
+  // peer (including same-process loopback, which uses the raw pointer).
+  // STARWAY_INBOX=0 disables.
+  static const bool inbox_on = [] {
+    const char* v = getenv("STARWAY_INBOX");
+    return !(v && !strcmp(v, "0"));
+  }();
+  if (inbox_on && gpu::available() && c->peer.has_gpu &&
+      memcmp(c->peer.host_id, host_id(), 16) == 0 && !c->inbox_l_active) {
+    std::string err;
+    if (gpu::inbox_create(&c->inbox_l, &err)) {
+      c->inbox_l_active = true;
+      enqueue_frame(c, FT_INBOX_OFFER, 0, 0, 0, &c->inbox_l,
+                    sizeof(gpu::InboxInfo), false);
+    } else {
+      SW_DBG("inbox create failed: %s", err.c_str());
+    }
+  }
 }
 
 // ---- eager path -----------------------------------------------------------
@@ -1431,10 +1495,8 @@ void Engine::begin_eager(Connection* c) {
   c->rx_msg_remaining = msg_len;
 
   // Match against posted recvs in FIFO order.
-  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
-    Op* r = *it;
-    if ((tag & r->tag_mask) == (r->tag & r->tag_mask)) {
-      posted_recvs_.erase(it);
+  if (Op* r = take_matching_recv(tag)) {
+    {
       if (msg_len > r->buf.size ||
           (r->buf.rows > 0 && msg_len != r->buf.size)) {
         // Truncation / strided-geometry mismatch: consume + fail
@@ -1458,7 +1520,6 @@ void Engine::begin_eager(Connection* c) {
       }
       c->rx_recv_op->recv_sender_tag = tag;
       c->rx_recv_op->recv_len = msg_len;
-      break;
     }
   }
   if (!c->rx_recv_op) {
@@ -1562,13 +1623,9 @@ void Engine::finish_eager_into_recv(Connection* c) {
 
 void Engine::handle_rts(Connection* c, const RtsDesc& rts, uint64_t tag,
                         uint64_t size, uint64_t sender_op) {
-  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
-    Op* r = *it;
-    if ((tag & r->tag_mask) == (r->tag & r->tag_mask)) {
-      posted_recvs_.erase(it);
-      start_gpu_pull(r, rts, tag, size, sender_op, c);
-      return;
-    }
+  if (Op* r = take_matching_recv(tag)) {
+    start_gpu_pull(r, rts, tag, size, sender_op, c);
+    return;
   }
   auto um = std::make_unique<UnexpectedMsg>();
   um->tag = tag;
@@ -1671,6 +1728,369 @@ void Engine::flush_small_pulls() {
   }
 }
 
+// ---- small-message inbox plane --------------------------------------------
+
+void Engine::inbox_release_seq(Connection* c, uint64_t seq) {
+  c->inbox_released.insert(seq);
+  while (c->inbox_released.count(c->inbox_consumed + 1)) {
+    c->inbox_released.erase(c->inbox_consumed + 1);
+    c->inbox_consumed++;
+  }
+  if (!c->dead &&
+      c->inbox_consumed - c->inbox_credited >=
+          std::max<uint64_t>(1, c->inbox_l.slots / 4)) {
+    enqueue_frame(c, FT_INBOX_CREDIT, 0, 0, c->inbox_consumed, nullptr, 0,
+                  true);
+    c->inbox_credited = c->inbox_consumed;
+  }
+}
+
+// Route one announced inbox message to a matched recv: batched local
+// unpack kernel (device dst) or pinned bounce (host / strided dst).
+void Engine::dispatch_smsg_to_recv(Connection* c, Op* r, uint64_t tag,
+                                   uint64_t size, uint64_t seq) {
+  if (size > r->buf.size || (r->buf.rows > 0 && size != r->buf.size)) {
+    fail_op(r, "receive failed: message truncated (len " +
+                   std::to_string(size) + " > buffer " +
+                   std::to_string(r->buf.size) + ")");
+    inbox_release_seq(c, seq);  // discard without reading the slot
+    return;
+  }
+  uint8_t* dst = nullptr;
+  if (r->buf.device >= 0 && r->buf.rows == 0 &&
+      r->buf.device == c->inbox_l.device)
+    dst = r->buf.ptr;
+  pending_unpacks_.push_back(PendingUnpack{c, seq, size, tag, r, dst});
+}
+
+void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
+                         uint64_t seq) {
+  c->inbox_seen_seq = seq;
+  // Doorbell copy that raced a disarm: this frame carries its tag.
+  if (armed_done_.active && armed_done_.conn == c && armed_done_.seq == seq) {
+    Op* r = armed_done_.recv_op;
+    armed_done_ = ArmedDone{};
+    stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+    stats_.bytes_received.fetch_add(size, std::memory_order_relaxed);
+    stats_.inbox_rx.fetch_add(1, std::memory_order_relaxed);
+    stats_.doorbell_rx.fetch_add(1, std::memory_order_relaxed);
+    Completion comp;
+    comp.kind = Completion::Kind::RecvDone;
+    comp.op = r;
+    comp.a = tag;
+    comp.b = size;
+    complete(std::move(comp));
+    inbox_release_seq(c, seq);
+    try_arm();
+    return;
+  }
+  // Doorbell fast path: a pre-armed kernel is watching exactly this seq.
+  if (armed_.ticket && armed_.conn == c && armed_.seq == seq) {
+    uint64_t sz = 0;
+    int st = gpu::arm_poll(armed_.ticket, &sz);
+    if (st == 0) {
+      // The payload and this control frame race each other; the kernel
+      // sees the slot within microseconds of the xGMI write landing.
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::milliseconds(
+                          (long)env_u64("STARWAY_ARM_WAIT_MS", 5));
+      while ((st = gpu::arm_poll(armed_.ticket, &sz)) == 0) {
+        if (std::chrono::steady_clock::now() > deadline) break;
+      }
+      if (st == 0) {
+        gpu::arm_cancel(armed_.ticket);
+        auto hard = std::chrono::steady_clock::now() +
+                    std::chrono::milliseconds(100);
+        while ((st = gpu::arm_poll(armed_.ticket, &sz)) == 0 &&
+               std::chrono::steady_clock::now() < hard)
+          sched_yield();
+      }
+    }
+    Op* r = armed_.recv_op;
+    gpu::arm_free(armed_.ticket);
+    armed_ = Armed{};
+    if (st == 1) {
+      for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it)
+        if (*it == r) {
+          posted_recvs_.erase(it);
+          break;
+        }
+      stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+      stats_.bytes_received.fetch_add(sz, std::memory_order_relaxed);
+      stats_.inbox_rx.fetch_add(1, std::memory_order_relaxed);
+      stats_.doorbell_rx.fetch_add(1, std::memory_order_relaxed);
+      Completion comp;
+      comp.kind = Completion::Kind::RecvDone;
+      comp.op = r;
+      comp.a = tag;
+      comp.b = sz;
+      complete(std::move(comp));
+      inbox_release_seq(c, seq);
+      try_arm();
+      return;
+    }
+    // nomatch / canceled / expired / stuck: the recv stays posted and the
+    // message goes through the ordinary path below.
+  }
+  if (Op* r = take_matching_recv(tag)) {
+    dispatch_smsg_to_recv(c, r, tag, size, seq);
+    return;
+  }
+  stats_.unexpected_rx.fetch_add(1, std::memory_order_relaxed);
+  auto um = std::make_unique<UnexpectedMsg>();
+  um->tag = tag;
+  um->size = size;
+  um->conn = c;
+  um->is_smsg = true;
+  um->smsg_seq = seq;
+  um->complete = true;
+  unexpected_.push_back(std::move(um));
+}
+
+void Engine::flush_pending_pushes() {
+  for (auto& [c, vec] : pending_pushes_) {
+    if (vec.empty()) continue;
+    if (c->dead) {
+      for (auto& pp : vec) {
+        gpu_sends_.erase(pp.op->id);
+        fail_op(pp.op, "send failed: connection reset");
+      }
+      vec.clear();
+      continue;
+    }
+    bool same_proc = memcmp(c->peer.uuid, process_uuid(), 16) == 0;
+    size_t i = 0;
+    while (i < vec.size()) {
+      // One launch per (destination ring, source device) batch.
+      int run_dev = vec[i].op->buf.device;
+      gpu::PushMsg msgs[32];
+      auto batch = std::make_unique<PushBatch>();
+      batch->conn = c;
+      int n = 0;
+      size_t j = i;
+      while (j < vec.size() && n < 32) {
+        if (vec[j].op->buf.device != run_dev) {
+          j++;
+          continue;
+        }
+        msgs[n] = gpu::PushMsg{vec[j].op->buf.ptr,
+                               (uint32_t)vec[j].op->buf.size, vec[j].seq,
+                               vec[j].op->tag};
+        batch->ops.push_back(vec[j].op);
+        vec.erase(vec.begin() + j);
+        n++;
+      }
+      std::string err;
+      batch->ticket =
+          gpu::inbox_push(c->inbox_r, same_proc, run_dev, msgs, n, &err);
+      if (!batch->ticket) {
+        // Push plane broken: fail these sends and stop using the inbox.
+        c->inbox_r_active = false;
+        for (Op* op : batch->ops) {
+          gpu_sends_.erase(op->id);
+          fail_op(op, "send failed: " + err);
+        }
+      } else {
+        push_batches_.push_back(std::move(batch));
+      }
+    }
+  }
+  pending_pushes_.clear();
+}
+
+void Engine::progress_pushes(bool& did_work) {
+  for (size_t i = 0; i < push_batches_.size();) {
+    PushBatch* b = push_batches_[i].get();
+    std::string err;
+    int r = gpu::push_poll(b->ticket, &err);
+    if (r == 0) {
+      i++;
+      continue;
+    }
+    did_work = true;
+    gpu::push_free(b->ticket);
+    for (Op* op : b->ops) {
+      gpu_sends_.erase(op->id);
+      if (r < 0) {
+        for (Op* f : pending_flushes_) f->flush_ops_pending.erase(op->id);
+        fail_op(op, "send failed: " + err);
+      } else if (!b->conn || b->conn->dead) {
+        for (Op* f : pending_flushes_) f->flush_ops_pending.erase(op->id);
+        fail_op(op, "send failed: connection reset");
+      } else {
+        on_send_wire_handoff(op, b->conn);
+        Completion comp;
+        comp.kind = Completion::Kind::SendDone;
+        comp.op = op;
+        complete(std::move(comp));
+      }
+    }
+    push_batches_.erase(push_batches_.begin() + i);
+  }
+}
+
+void Engine::flush_pending_unpacks() {
+  while (!pending_unpacks_.empty()) {
+    Connection* c = pending_unpacks_[0].conn;
+    auto batch = std::make_unique<UnpackBatch>();
+    batch->conn = c;
+    gpu::UnpackMsg msgs[32];
+    int n = 0;
+    for (size_t i = 0; i < pending_unpacks_.size() && n < 32;) {
+      if (pending_unpacks_[i].conn == c) {
+        msgs[n] = gpu::UnpackMsg{pending_unpacks_[i].seq,
+                                 (uint32_t)pending_unpacks_[i].size,
+                                 pending_unpacks_[i].dst};
+        batch->msgs.push_back(pending_unpacks_[i]);
+        pending_unpacks_.erase(pending_unpacks_.begin() + i);
+        n++;
+      } else {
+        i++;
+      }
+    }
+    std::string err;
+    batch->ticket = gpu::inbox_unpack(c->inbox_l, msgs, n, &err);
+    if (!batch->ticket) {
+      for (auto& m : batch->msgs) {
+        fail_op(m.recv_op, "receive failed: " + err);
+        inbox_release_seq(c, m.seq);
+      }
+      continue;
+    }
+    batch->done.assign(batch->msgs.size(), false);
+    batch->remaining = batch->msgs.size();
+    unpack_batches_.push_back(std::move(batch));
+  }
+}
+
+void Engine::progress_unpacks(bool& did_work) {
+  for (size_t i = 0; i < unpack_batches_.size();) {
+    UnpackBatch* b = unpack_batches_[i].get();
+    for (size_t k = 0; k < b->msgs.size(); k++) {
+      if (b->done[k]) continue;
+      std::string err;
+      int r = gpu::unpack_poll(b->ticket, (int)k, &err);
+      if (r == 0) continue;
+      did_work = true;
+      b->done[k] = true;
+      b->remaining--;
+      PendingUnpack& m = b->msgs[k];
+      if (r < 0) {
+        // Payload never landed (sender died mid-push): the message is
+        // lost; the recv stays pending (unflushed-close contract).
+        repost_recv_front(m.recv_op);
+        inbox_release_seq(m.conn, m.seq);
+        continue;
+      }
+      if (!m.dst) {
+        const uint8_t* bounce = gpu::unpack_bounce(b->ticket, (int)k);
+        if (m.recv_op->buf.device < 0) {
+          memcpy(m.recv_op->buf.ptr, bounce, m.size);
+        } else {
+          // Strided / other-device recv: re-upload from a heap copy via
+          // the ordinary h2d machinery (rare path, <= slot capacity).
+          RawBuf heap;
+          std::string err2;
+          if (!heap.alloc(m.size)) {
+            fail_op(m.recv_op, "receive failed: bounce allocation failed");
+            inbox_release_seq(m.conn, m.seq);
+            continue;
+          }
+          memcpy(heap.data(), bounce, m.size);
+          void* ticket =
+              gpu::begin_h2d(m.recv_op->buf, heap.data(), m.size, &err2);
+          if (!ticket) {
+            fail_op(m.recv_op, "receive failed: " + err2);
+            inbox_release_seq(m.conn, m.seq);
+            continue;
+          }
+          auto pull = std::make_unique<GpuPull>();
+          pull->ticket = ticket;
+          pull->recv_op = m.recv_op;
+          pull->conn = nullptr;
+          pull->sender_op_id = 0;
+          pull->tag = m.tag;
+          pull->len = m.size;
+          gpu::attach_bounce(ticket, std::move(heap));
+          gpu_pulls_.push_back(std::move(pull));
+          inbox_release_seq(m.conn, m.seq);
+          continue;
+        }
+      }
+      stats_.msgs_received.fetch_add(1, std::memory_order_relaxed);
+      stats_.bytes_received.fetch_add(m.size, std::memory_order_relaxed);
+      stats_.inbox_rx.fetch_add(1, std::memory_order_relaxed);
+      Completion comp;
+      comp.kind = Completion::Kind::RecvDone;
+      comp.op = m.recv_op;
+      comp.a = m.tag;
+      comp.b = m.size;
+      complete(std::move(comp));
+      inbox_release_seq(m.conn, m.seq);
+    }
+    if (b->remaining == 0) {
+      gpu::unpack_free(b->ticket);
+      unpack_batches_.erase(unpack_batches_.begin() + i);
+    } else {
+      i++;
+    }
+  }
+}
+
+void Engine::try_arm() {
+  static const bool arm_on = [] {
+    const char* v = getenv("STARWAY_DOORBELL");
+    return !(v && !strcmp(v, "0"));
+  }();
+  if (!arm_on || armed_.ticket || armed_done_.active) return;
+  if (posted_recvs_.empty()) return;
+  Op* r = posted_recvs_.front();
+  if (r->buf.device < 0 || r->buf.rows > 0 || r->buf.size == 0) return;
+  // Only safe with exactly one live connection (its inbox FIFO is then the
+  // only source of inbox messages that could race the doorbell).
+  Connection* target = nullptr;
+  int n_live = 0;
+  for (auto& c : conns_) {
+    if (c->dead) continue;
+    n_live++;
+    target = c.get();
+  }
+  if (n_live != 1 || !target || !target->inbox_l_active) return;
+  if (r->buf.device != target->inbox_l.device) return;
+  // Only worth a resident kernel when the expected message can actually
+  // ride the inbox (small buffer): large recvs are served by the RTS pull
+  // path and would just churn expired doorbells.
+  if (r->buf.size > target->inbox_l.slot_bytes - gpu::kInboxHdrBytes) return;
+  // Never skip ahead of inbox messages already being delivered.
+  if (!pending_unpacks_.empty() || !unpack_batches_.empty()) return;
+  for (auto& um : unexpected_)
+    if (um->is_smsg && um->conn == target) return;
+  std::string err;
+  void* t = gpu::arm_recv(target->inbox_l, target->inbox_seen_seq + 1,
+                          r->tag, r->tag_mask, r->buf.ptr, r->buf.size,
+                          &err);
+  if (!t) {
+    SW_DBG("arm failed: %s", err.c_str());
+    return;
+  }
+  armed_ = Armed{t, r, target, target->inbox_seen_seq + 1};
+}
+
+void Engine::progress_armed(bool& did_work) {
+  if (!armed_.ticket) return;
+  uint64_t sz = 0;
+  int st = gpu::arm_poll(armed_.ticket, &sz);
+  if (st == 4) {
+    // Bounded wait expired with no message: re-arm at the same sequence.
+    did_work = true;
+    gpu::arm_free(armed_.ticket);
+    armed_ = Armed{};
+    try_arm();
+  }
+  // 1 (copied) / 2 (nomatch) are resolved by handle_smsg when the control
+  // frame lands; 0 keeps waiting.
+}
+
 void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
                             uint64_t size, uint64_t sender_op,
                             Connection* c) {
@@ -1693,7 +2113,7 @@ void Engine::start_cma_pull(Op* recv_op, const CmaDesc& cma, uint64_t tag,
       std::string err = "cma unavailable (bounce allocation failed)";
       enqueue_frame(c, FT_RECV_FAIL, 0, sender_op, 0, err.data(), err.size(),
                     true);
-      posted_recvs_.push_front(recv_op);
+      repost_recv_front(recv_op);
       return;
     }
   }
@@ -1766,7 +2186,7 @@ void Engine::progress_cma(bool& did_work) {
   for (size_t i = 0; i < cma_pulls_.size();) {
     CmaPull* p = cma_pulls_[i].get();
     if (p->conn && p->conn->dead) {
-      posted_recvs_.push_front(p->recv_op);  // data lost; recv stays pending
+      repost_recv_front(p->recv_op);  // data lost; recv stays pending
       cma_pulls_.erase(cma_pulls_.begin() + i);
       continue;
     }
@@ -1798,12 +2218,12 @@ void Engine::progress_cma(bool& did_work) {
             "cma unavailable (" + std::string(strerror(errno)) + ")";
         enqueue_frame(p->conn, FT_RECV_FAIL, 0, p->sender_op_id, 0,
                       err.data(), err.size(), true);
-        posted_recvs_.push_front(p->recv_op);
+        repost_recv_front(p->recv_op);
       } else {
         // ESRCH/EFAULT: sender died or freed the buffer mid-pull — the
         // message is lost; the recv stays pending (unflushed-close
         // contract).
-        posted_recvs_.push_front(p->recv_op);
+        repost_recv_front(p->recv_op);
       }
       cma_pulls_.erase(cma_pulls_.begin() + i);
       continue;
@@ -1979,6 +2399,76 @@ void Engine::unstage_unexp(UnexpectedMsg* um) {
   um->staged = 0;
 }
 
+// Resolve a doorbell conflict: another delivery path selected the armed
+// recv. Cancel the wait kernel and see who won. Returns true when the recv
+// is still available; false when the kernel consumed a message into it
+// first (armed_done_ records the copy until its SMSG frame arrives).
+bool Engine::steal_armed(Op* r) {
+  gpu::arm_cancel(armed_.ticket);
+  uint64_t sz = 0;
+  int st;
+  auto deadline =
+      std::chrono::steady_clock::now() + std::chrono::milliseconds(100);
+  while ((st = gpu::arm_poll(armed_.ticket, &sz)) == 0) {
+    if (std::chrono::steady_clock::now() > deadline) break;
+    sched_yield();
+  }
+  bool copied = (st == 1);
+  gpu::arm_free(armed_.ticket);
+  Connection* conn = armed_.conn;
+  uint64_t seq = armed_.seq;
+  armed_ = Armed{};
+  if (!copied) return true;
+  // The kernel copied message `seq` into r before we could cancel: r is
+  // spoken for; complete it when the control frame (with the tag) lands.
+  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
+    if (*it == r) {
+      posted_recvs_.erase(it);
+      break;
+    }
+  }
+  armed_done_ = ArmedDone{true, conn, seq, sz, r};
+  return false;
+}
+
+// Find and REMOVE the first posted recv matching `tag`, resolving doorbell
+// conflicts. Every delivery path matches through here so the doorbell can
+// never race a completion.
+Op* Engine::take_matching_recv(uint64_t tag) {
+restart:
+  for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it) {
+    Op* r = *it;
+    if ((tag & r->tag_mask) != (r->tag & r->tag_mask)) continue;
+    if (armed_.ticket && armed_.recv_op == r) {
+      if (!steal_armed(r)) goto restart;  // doorbell consumed r: rescan
+      // steal_armed left r posted; the iterator may be stale — rescan.
+      for (auto it2 = posted_recvs_.begin(); it2 != posted_recvs_.end();
+           ++it2) {
+        if (*it2 == r) {
+          posted_recvs_.erase(it2);
+          return r;
+        }
+      }
+      goto restart;
+    }
+    posted_recvs_.erase(it);
+    return r;
+  }
+  return nullptr;
+}
+
+void Engine::repost_recv_front(Op* r) {
+  if (armed_.ticket) {
+    // The front of the posted queue is changing: the armed recv is no
+    // longer the first match candidate, so the doorbell must stand down.
+    Op* armed_r = armed_.recv_op;
+    if (!steal_armed(armed_r)) {
+      // It consumed a message meanwhile; armed_done_ tracks it.
+    }
+  }
+  posted_recvs_.push_front(r);
+}
+
 void Engine::match_or_stash_recv(Op* op) {
   if (!try_match_unexpected(op)) posted_recvs_.push_back(op);
 }
@@ -2002,6 +2492,13 @@ bool Engine::try_match_unexpected(Op* op) {
       Connection* c = um->conn;
       unexpected_.erase(it);
       start_cma_pull(op, cma, tag, size, sop, c);
+      return true;
+    }
+    if (um->is_smsg) {
+      Connection* c = um->conn;
+      uint64_t tag = um->tag, size = um->size, seq = um->smsg_seq;
+      unexpected_.erase(it);
+      dispatch_smsg_to_recv(c, op, tag, size, seq);
       return true;
     }
     if (!um->complete) {
@@ -2263,6 +2760,31 @@ void Engine::check_flush_progress() {
 void Engine::on_conn_dead(Connection* c) {
   if (c->dead) return;
   c->dead = true;
+  // Doorbell armed on this connection: stand down (a copy that already
+  // happened has no control frame coming — the recv fails below).
+  if (armed_.ticket && armed_.conn == c) {
+    Op* r = armed_.recv_op;
+    if (!steal_armed(r)) {
+      // Copied, but the tag-bearing frame is lost with the connection.
+      armed_done_ = ArmedDone{};
+      fail_op(r, "receive failed: connection reset");
+    }
+  }
+  if (armed_done_.active && armed_done_.conn == c) {
+    fail_op(armed_done_.recv_op, "receive failed: connection reset");
+    armed_done_ = ArmedDone{};
+  }
+  // Un-launched inbox pushes to this connection never left the building.
+  std::vector<uint64_t> dead_send_ids;
+  auto pp = pending_pushes_.find(c);
+  if (pp != pending_pushes_.end()) {
+    for (auto& p : pp->second) {
+      dead_send_ids.push_back(p.op->id);
+      gpu_sends_.erase(p.op->id);
+      fail_op(p.op, "send failed: connection reset");
+    }
+    pending_pushes_.erase(pp);
+  }
   if (c->fd >= 0) {
     ::close(c->fd);
     c->fd = -1;
@@ -2273,14 +2795,14 @@ void Engine::on_conn_dead(Connection* c) {
   // A matched recv whose message was mid-stream goes back to pending: the
   // data is lost but the recv must NOT complete (flush-semantics tests).
   if (c->rx_recv_op) {
-    posted_recvs_.push_front(c->rx_recv_op);
+    repost_recv_front(c->rx_recv_op);
     c->rx_recv_op = nullptr;
   }
   if (c->rx_unexp) {
     // Incomplete unexpected message: drop it.
     for (auto it = unexpected_.begin(); it != unexpected_.end(); ++it) {
       if (it->get() == c->rx_unexp) {
-        if ((*it)->bound_recv) posted_recvs_.push_front((*it)->bound_recv);
+        if ((*it)->bound_recv) repost_recv_front((*it)->bound_recv);
         unstage_unexp(it->get());
         unexpected_.erase(it);
         break;
@@ -2334,10 +2856,9 @@ void Engine::on_conn_dead(Connection* c) {
     }
   }
   // GPU sends routed to this conn will never be acked. Ops owned by the
-  // d2h staging list are only unregistered here — progress_d2h reaps them
-  // (double-delete hazard otherwise). Collect the op ids so flushes that
-  // snapshotted them fail instead of hanging forever.
-  std::vector<uint64_t> dead_send_ids;
+  // d2h/push staging lists are only unregistered here — their machinery
+  // reaps them (double-delete hazard otherwise). Collect the op ids so
+  // flushes that snapshotted them fail instead of hanging forever.
   for (auto it = gpu_sends_.begin(); it != gpu_sends_.end();) {
     if (it->second->conn == c) {
       dead_send_ids.push_back(it->first);
@@ -2438,6 +2959,53 @@ void Engine::teardown() {
     std::vector<Op*> cmds;
     drain_commands(cmds);
     for (Op* op : cmds) fail_op(op, "operation canceled (endpoint closing)");
+  }
+  // 1b. Small-message inbox wind-down: stand the doorbell down, launch any
+  //     pending pushes (payload capture for already-completed... no — push
+  //     ops have NOT completed; they are canceled below if their kernel
+  //     cannot finish), and let in-flight push/unpack kernels drain
+  //     (bounded; they are microsecond-scale copies).
+  if (armed_.ticket) {
+    Op* r = armed_.recv_op;
+    steal_armed(r);
+    // If the doorbell copied at the last instant, armed_done_ now owns the
+    // recv (erased from posted_recvs_) and is canceled just below; else r
+    // stays posted and step 4 cancels it.
+  }
+  if (armed_done_.active) {
+    fail_op(armed_done_.recv_op, "operation canceled (endpoint closing)");
+    armed_done_ = ArmedDone{};
+  }
+  for (auto& m : pending_unpacks_)
+    fail_op(m.recv_op, "operation canceled (endpoint closing)");
+  pending_unpacks_.clear();
+  if (!pending_pushes_.empty()) flush_pending_pushes();
+  {
+    auto sm_deadline =
+        std::chrono::steady_clock::now() + std::chrono::seconds(5);
+    while ((!push_batches_.empty() || !unpack_batches_.empty()) &&
+           std::chrono::steady_clock::now() < sm_deadline) {
+      bool did = false;
+      progress_pushes(did);
+      progress_unpacks(did);
+      if (!did) sched_yield();
+    }
+    for (auto& b : push_batches_) {
+      gpu::push_free(b->ticket);
+      for (Op* op : b->ops) {
+        gpu_sends_.erase(op->id);
+        fail_op(op, "operation canceled (endpoint closing)");
+      }
+    }
+    push_batches_.clear();
+    for (auto& b : unpack_batches_) {
+      gpu::unpack_free(b->ticket);
+      for (size_t k = 0; k < b->msgs.size(); k++)
+        if (!b->done[k])
+          fail_op(b->msgs[k].recv_op,
+                  "operation canceled (endpoint closing)");
+    }
+    unpack_batches_.clear();
   }
   // 2. Launch any still-pending batched pulls, then let in-flight GPU
   //    pulls finish (bounded; they are plain copies), complete them and
@@ -2556,6 +3124,13 @@ void Engine::teardown() {
     c->txq.clear();
     c->shm_rx = false;
     c->shm.reset();
+    if (c->inbox_l_active) {
+      // Freed only after the close drain: a peer's in-flight push kernel
+      // finishes in microseconds, and no new pushes can start once the
+      // connection is down.
+      gpu::inbox_destroy(c->inbox_l);
+      c->inbox_l_active = false;
+    }
   }
   if (listen_fd_ >= 0) {
     ::close(listen_fd_);

@@ -170,12 +170,16 @@ struct ExportEntry {
 };
 static std::map<void*, ExportEntry> g_export_cache;
 
-// Import cache: (device, 64B handle) -> mapped base.
+// Import cache: (device, 64B handle) -> mapped base. Guarded by its own
+// mutex because it is shared between the pull path (g_mu domain) and the
+// small-message inbox path (sm_mu domain in smallmsg.hip).
 using HandleKey = std::array<uint8_t, kIpcHandleBytes>;
+static std::mutex g_ipc_mu;
 static std::map<std::pair<int, HandleKey>, void*> g_import_cache;
 
 void ipc_close_all() {
-  std::lock_guard<std::mutex> lk(g_mu);
+  std::lock_guard<std::mutex> lk0(g_mu);
+  std::lock_guard<std::mutex> lk(g_ipc_mu);
   int prev = 0;
   hipGetDevice(&prev);
   for (auto& [key, base] : g_import_cache) {
@@ -269,6 +273,31 @@ static void pool_put_event(int device, hipEvent_t ev) {
   }
 }
 
+// Map a peer's exported allocation on open_device (cached per device ×
+// handle). Shared by the pull path and the small-message inbox push path.
+void* import_ipc(const uint8_t* handle, int open_device, std::string* err) {
+  std::lock_guard<std::mutex> lk(g_ipc_mu);
+  HandleKey key;
+  memcpy(key.data(), handle, kIpcHandleBytes);
+  auto ck = std::make_pair(open_device, key);
+  auto it = g_import_cache.find(ck);
+  if (it != g_import_cache.end()) return it->second;
+  hipIpcMemHandle_t h;
+  memcpy(&h, handle, kIpcHandleBytes);
+  void* base = nullptr;
+  int prev;
+  hipGetDevice(&prev);
+  hipSetDevice(open_device);
+  hipError_t e = hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess);
+  hipSetDevice(prev);
+  if (e != hipSuccess) {
+    *err = std::string("hipIpcOpenMemHandle: ") + hipGetErrorString(e);
+    return nullptr;
+  }
+  g_import_cache[ck] = base;
+  return base;
+}
+
 static void* resolve_src(const RtsDesc& rts, int open_device,
                          std::string* err) {
   bool same_proc = memcmp(rts.src_uuid, process_uuid(), 16) == 0;
@@ -277,28 +306,8 @@ static void* resolve_src(const RtsDesc& rts, int open_device,
     *err = "peer did not export an IPC handle (cross-process GPU transfer)";
     return nullptr;
   }
-  HandleKey key;
-  memcpy(key.data(), rts.ipc_handle, kIpcHandleBytes);
-  auto ck = std::make_pair(open_device, key);
-  auto it = g_import_cache.find(ck);
-  void* base = nullptr;
-  if (it != g_import_cache.end()) {
-    base = it->second;
-  } else {
-    hipIpcMemHandle_t h;
-    memcpy(&h, rts.ipc_handle, kIpcHandleBytes);
-    int prev;
-    hipGetDevice(&prev);
-    hipSetDevice(open_device);
-    hipError_t e =
-        hipIpcOpenMemHandle(&base, h, hipIpcMemLazyEnablePeerAccess);
-    hipSetDevice(prev);
-    if (e != hipSuccess) {
-      *err = std::string("hipIpcOpenMemHandle: ") + hipGetErrorString(e);
-      return nullptr;
-    }
-    g_import_cache[ck] = base;
-  }
+  void* base = import_ipc(rts.ipc_handle, open_device, err);
+  if (!base) return nullptr;
   return (uint8_t*)base + rts.offset;
 }
 

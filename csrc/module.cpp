@@ -197,6 +197,9 @@ PYBIND11_MODULE(_core, m) {
              d["unexpected_rx"] = st.unexpected_rx.load();
              d["unexp_staged_bytes"] = st.unexp_staged_bytes.load();
              d["deferred_sends"] = st.deferred_sends.load();
+             d["inbox_rx"] = st.inbox_rx.load();
+             d["inbox_tx"] = st.inbox_tx.load();
+             d["doorbell_rx"] = st.doorbell_rx.load();
              return d;
            })
       .def("list_clients",
@@ -271,6 +274,9 @@ PYBIND11_MODULE(_core, m) {
              d["unexpected_rx"] = st.unexpected_rx.load();
              d["unexp_staged_bytes"] = st.unexp_staged_bytes.load();
              d["deferred_sends"] = st.deferred_sends.load();
+             d["inbox_rx"] = st.inbox_rx.load();
+             d["inbox_tx"] = st.inbox_tx.load();
+             d["doorbell_rx"] = st.doorbell_rx.load();
              return d;
            })
       .def("evaluate_perf", [](PyClient& c, uint64_t msg_size) {
