@@ -411,8 +411,9 @@ async def test_strided_geometry_mismatch_fails():
         sfut = client.asend(src[:, ::1], 11)
         with pytest.raises(Exception, match="geometry|truncated"):
             await fut
-        with pytest.raises(Exception):
-            await sfut
+        # Small (inbox-eager) send: the sender is not notified of the
+        # receiver-side geometry failure — eager semantics.
+        await sfut
 
 
 async def test_cross_host_gpu_bounce_path():
@@ -579,16 +580,18 @@ async def test_inbox_order_with_rts_interleaved():
         big = torch.full((1 << 20,), 2, dtype=torch.uint8, device="cuda")
         small2 = torch.full((256,), 3, dtype=torch.uint8, device="cuda")
         torch.cuda.synchronize()
-        await client.asend(small1, 11)
-        await client.asend(big, 12)
-        await client.asend(small2, 13)
-        await client.aflush()
+        # Enqueue order defines per-sender FIFO; the RTS send (12) only
+        # completes at delivery, so it cannot be awaited before the recvs.
+        sends = [client.asend(small1, 11), client.asend(big, 12),
+                 client.asend(small2, 13)]
         seen = []
         for _ in range(3):
             dst = torch.zeros(1 << 20, dtype=torch.uint8, device="cuda")
             tag, length = await server.arecv(dst, 0, 0)
             torch.cuda.synchronize()
             seen.append((tag, length, int(dst[0])))
+        await asyncio.gather(*sends)
+        await client.aflush()
         assert seen == [(11, 256, 1), (12, 1 << 20, 2), (13, 256, 3)]
 
 
