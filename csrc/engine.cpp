@@ -2043,7 +2043,10 @@ void Engine::try_arm() {
     return !(v && !strcmp(v, "0"));
   }();
   if (!arm_on || armed_.ticket || armed_done_.active) return;
-  if (posted_recvs_.empty()) return;
+  // Latency pattern only: with several recvs outstanding (throughput
+  // pattern) a doorbell would serialize the batched unpack path into one
+  // arm launch + inline wait per message.
+  if (posted_recvs_.size() != 1) return;
   Op* r = posted_recvs_.front();
   if (r->buf.device < 0 || r->buf.rows > 0 || r->buf.size == 0) return;
   // Only safe with exactly one live connection (its inbox FIFO is then the

@@ -18,6 +18,8 @@ from __future__ import annotations
 import asyncio
 import ctypes
 import os
+import weakref
+from collections import deque
 from collections.abc import Callable
 from importlib.util import find_spec
 from pathlib import Path
@@ -197,10 +199,10 @@ class Server:
         ret: asyncio.Future[None] = asyncio.Future(loop=loop)
 
         def ok() -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+            _disp(ret.get_loop()).post(_set_result, ret, None)
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._server.send(client_ep, _norm_buffer(buffer), tag, ok, bad)
         return ret
@@ -212,11 +214,10 @@ class Server:
         ret: asyncio.Future[tuple[int, int]] = asyncio.Future(loop=loop)
 
         def ok(sender_tag: int, length: int) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret,
-                                                (sender_tag, length))
+            _disp(ret.get_loop()).post(_set_result, ret, (sender_tag, length))
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._server.recv(_norm_buffer(buffer), tag, tag_mask, ok, bad)
         return ret
@@ -227,10 +228,10 @@ class Server:
         ret: asyncio.Future[None] = asyncio.Future(loop=loop)
 
         def ok() -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+            _disp(ret.get_loop()).post(_set_result, ret, None)
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._server.flush(ok, bad)
         return ret
@@ -242,10 +243,10 @@ class Server:
         ret: asyncio.Future[None] = asyncio.Future(loop=loop)
 
         def ok() -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+            _disp(ret.get_loop()).post(_set_result, ret, None)
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._server.flush_ep(client_ep, ok, bad)
         return ret
@@ -326,10 +327,10 @@ class Client:
         ret: asyncio.Future[None] = asyncio.Future(loop=loop)
 
         def ok() -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+            _disp(ret.get_loop()).post(_set_result, ret, None)
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._client.send(_norm_buffer(buffer), tag, ok, bad)
         return ret
@@ -341,11 +342,10 @@ class Client:
         ret: asyncio.Future[tuple[int, int]] = asyncio.Future(loop=loop)
 
         def ok(sender_tag: int, length: int) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret,
-                                                (sender_tag, length))
+            _disp(ret.get_loop()).post(_set_result, ret, (sender_tag, length))
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._client.recv(_norm_buffer(buffer), tag, tag_mask, ok, bad)
         return ret
@@ -356,16 +356,59 @@ class Client:
         ret: asyncio.Future[None] = asyncio.Future(loop=loop)
 
         def ok() -> None:
-            ret.get_loop().call_soon_threadsafe(_set_result, ret, None)
+            _disp(ret.get_loop()).post(_set_result, ret, None)
 
         def bad(reason: str) -> None:
-            ret.get_loop().call_soon_threadsafe(_set_exception, ret, reason)
+            _disp(ret.get_loop()).post(_set_exception, ret, reason)
 
         self._client.flush(ok, bad)
         return ret
 
     def evaluate_perf(self, msg_size: int) -> float:
         return self._client.evaluate_perf(msg_size)
+
+
+
+class _LoopDispatcher:
+    """Coalesced cross-thread completion delivery.
+
+    The engine thread fires op callbacks under the GIL; scheduling each one
+    with loop.call_soon_threadsafe costs a self-pipe write + one loop
+    callback PER OP. Under a completion burst (64 concurrent small
+    messages) this dispatcher queues the completions and schedules ONE
+    wakeup for the whole batch. All state is GIL-serialized (both the
+    engine thread and the loop thread hold the GIL when touching it).
+    """
+
+    __slots__ = ("_loop", "_q", "_armed")
+
+    def __init__(self, loop: asyncio.AbstractEventLoop) -> None:
+        self._loop = loop
+        self._q: deque = deque()
+        self._armed = False
+
+    def post(self, fn, *args) -> None:
+        self._q.append((fn, args))
+        if not self._armed:
+            self._armed = True
+            self._loop.call_soon_threadsafe(self._drain)
+
+    def _drain(self) -> None:
+        self._armed = False
+        q = self._q
+        while q:
+            fn, args = q.popleft()
+            fn(*args)
+
+
+_dispatchers: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
+
+
+def _disp(loop: asyncio.AbstractEventLoop) -> _LoopDispatcher:
+    d = _dispatchers.get(loop)
+    if d is None:
+        d = _dispatchers[loop] = _LoopDispatcher(loop)
+    return d
 
 
 def _set_result(fut: asyncio.Future, value) -> None:
