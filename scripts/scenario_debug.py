@@ -6,6 +6,7 @@ from __future__ import annotations
 
 import argparse
 import asyncio
+import gc
 import json
 import sys
 import time
@@ -73,7 +74,11 @@ def main():
     p.add_argument("--nbytes", type=int, default=1024)
     p.add_argument("--window", type=int, default=64)
     p.add_argument("--batches", type=int, default=10)
+    p.add_argument("--gc-off", action="store_true")
     args = p.parse_args()
+    if args.gc_off:
+        gc.disable()
+        gc.freeze()
     asyncio.run(run(args))
 
 

@@ -124,6 +124,10 @@ class _CudaShim:
                 "(call .contiguous() first)"
             )
         self._t = t
+        try:
+            t._sw_cai = self.__cuda_array_interface__
+        except Exception:
+            pass  # tensor subclass without attribute support
 
 
 def _norm_buffer(buf: Any) -> Any:
@@ -136,7 +140,17 @@ def _norm_buffer(buf: Any) -> Any:
     mod = type(buf).__module__
     if mod.startswith("torch"):
         if buf.is_cuda:
-            return _CudaShim(buf)
+            # The interface dict is cached on the tensor (hot benches reuse
+            # message buffers); the shim itself is rebuilt per op because
+            # it doubles as the op's keepalive (a cached shim would form a
+            # tensor<->shim cycle and pin GPU memory on the GC).
+            cai = getattr(buf, "_sw_cai", None)
+            if cai is None or cai["data"][0] != buf.data_ptr():
+                return _CudaShim(buf)
+            shim = _CudaShim.__new__(_CudaShim)
+            shim.__cuda_array_interface__ = cai
+            shim._t = buf
+            return shim
         return buf.numpy()
     return buf
 
