@@ -499,6 +499,10 @@ class Engine {
  public:
   Mode mode_;
   std::atomic<int> status_{0};
+  // HIP device the inbox ring (and other engine-owned device state) lands
+  // on: captured from the CONSTRUCTING thread's current device (the
+  // engine's own thread never calls hipSetDevice). -1 = no GPU.
+  int preferred_device_ = -1;
 
  private:
   // Command queue: Python threads -> engine thread.
@@ -651,6 +655,7 @@ class Engine {
 namespace gpu {
 bool available();
 int device_count();
+int current_device();  // calling thread's device, -1 without GPU
 // Fill an RtsDesc for a device buffer (ipc handle or raw ptr).
 bool make_rts(const BufferRef& buf, RtsDesc* out, std::string* err);
 // Begin an async pull of `size` bytes described by `rts` into dst (host or
@@ -686,7 +691,7 @@ void ipc_close_all();
 // kernel on the receiver. See smallmsg.hip header comment for the design;
 // InboxInfo / kInboxHdrBytes are declared above (wire ABI).
 // ---------------------------------------------------------------------------
-bool inbox_create(InboxInfo* out, std::string* err);
+bool inbox_create(InboxInfo* out, int device, std::string* err);
 void inbox_destroy(const InboxInfo& ib);
 
 struct PushMsg {

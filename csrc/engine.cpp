@@ -1473,7 +1473,7 @@ void Engine::on_hello(Connection* c) {
   if (inbox_on && gpu::available() && c->peer.has_gpu &&
       memcmp(c->peer.host_id, host_id(), 16) == 0 && !c->inbox_l_active) {
     std::string err;
-    if (gpu::inbox_create(&c->inbox_l, &err)) {
+    if (gpu::inbox_create(&c->inbox_l, preferred_device_, &err)) {
       c->inbox_l_active = true;
       enqueue_frame(c, FT_INBOX_OFFER, 0, 0, 0, &c->inbox_l,
                     sizeof(gpu::InboxInfo), false);

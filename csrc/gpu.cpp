@@ -47,6 +47,14 @@ static int cached_device_count() {
 }
 
 bool available() { return cached_device_count() > 0; }
+
+// Current device of the CALLING thread (Python thread at object
+// construction), used to place engine-owned device state.
+int current_device() {
+  if (!available()) return -1;
+  int d = 0;
+  return hipGetDevice(&d) == hipSuccess ? d : -1;
+}
 int device_count() { return cached_device_count(); }
 
 // Optional roctx range markers (STARWAY_ROCTX=1): rocprofv3 --marker-trace

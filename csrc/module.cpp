@@ -87,12 +87,16 @@ static BufferRef parse_buffer(py::handle obj, bool writable) {
 
 struct PyServer {
   Engine engine{Engine::ServerMode};
-  explicit PyServer(Context&) {}
+  explicit PyServer(Context&) {
+    engine.preferred_device_ = gpu::current_device();
+  }
 };
 
 struct PyClient {
   Engine engine{Engine::ClientMode};
-  explicit PyClient(Context&) {}
+  explicit PyClient(Context&) {
+    engine.preferred_device_ = gpu::current_device();
+  }
 };
 
 }  // namespace sw

@@ -295,10 +295,16 @@ static hipError_t sm_event(int device, hipEvent_t* ev) {
 // inbox lifecycle
 // ---------------------------------------------------------------------------
 
-bool inbox_create(InboxInfo* out, std::string* err) {
+bool inbox_create(InboxInfo* out, int device, std::string* err) {
   std::lock_guard<std::mutex> lk(sm_mu);
-  int dev = 0;
-  if (hipGetDevice(&dev) != hipSuccess) {
+  // The engine's progress thread never calls hipSetDevice, so its current
+  // device is always 0 — the ring must land on the device the USER works
+  // on (captured at Server/Client construction on the Python thread; one
+  // process per GPU in the bench/production topology).
+  int prev = 0;
+  hipGetDevice(&prev);
+  int dev = device >= 0 ? device : prev;
+  if (hipSetDevice(dev) != hipSuccess) {
     *err = "no HIP device";
     return false;
   }
@@ -309,6 +315,7 @@ bool inbox_create(InboxInfo* out, std::string* err) {
   void* base = nullptr;
   hipError_t e = hipMalloc(&base, (size_t)slots * slot_bytes);
   if (e != hipSuccess) {
+    hipSetDevice(prev);
     *err = std::string("inbox alloc: ") + hipGetErrorString(e);
     return false;
   }
@@ -316,6 +323,7 @@ bool inbox_create(InboxInfo* out, std::string* err) {
   hipDeviceSynchronize();  // seq words must read 0 before the export leaks
   hipIpcMemHandle_t h;
   e = hipIpcGetMemHandle(&h, base);
+  hipSetDevice(prev);
   if (e != hipSuccess) {
     hipFree(base);
     *err = std::string("inbox export: ") + hipGetErrorString(e);

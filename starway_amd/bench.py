@@ -10,6 +10,7 @@ numpy buffers vs HIP device tensors (the reference had no GPU path), and
 from __future__ import annotations
 
 import argparse
+import gc
 import asyncio
 import json
 import os
@@ -226,7 +227,16 @@ async def run_client(args: argparse.Namespace) -> list[ScenarioResult]:
             print(f"[client] Starting '{name}' with {overrides or 'defaults'}.")
             await session.send_control({"scenario": name, "config": overrides})
             await session.wait_ready()
-            result = await scenario.client_runner(context, overrides)
+            # Bench hygiene: a gen-2 GC pause (~40 ms, triggered by the
+            # future/closure churn of concurrent-op scenarios) would
+            # dominate a batch sample; collect up front, disable during
+            # the timed region.
+            gc.collect()
+            gc.disable()
+            try:
+                result = await scenario.client_runner(context, overrides)
+            finally:
+                gc.enable()
             results.append(result)
             await session.wait_done()
             print(f"[client] Completed '{name}'.")
