@@ -733,9 +733,12 @@ void free_ticket(void* ticket);
 void synchronize_all();
 bool copy_device_sync(void* dst, const void* src, size_t n, int device,
                       std::string* err);
-// Stats for evaluate_perf / transports introspection.
+// Stats for evaluate_perf / transports introspection. Values come from a
+// cached on-box copy microbenchmark (~/.cache/starway/perf.cal) with
+// round-1 measurements as analytic fallback.
 double same_gpu_copy_gbps();
 double xgmi_link_gbps();
+bool calibrate(bool force, std::string* err);  // run + persist microbench
 }  // namespace gpu
 
 }  // namespace sw

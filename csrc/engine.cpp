@@ -505,10 +505,12 @@ double Engine::perf_model(bool peer_gpu, bool same_proc,
   bool self_gpu = gpu::available();
   double lat, bw;
   if (self_gpu && peer_gpu) {
-    lat = 33e-6;  // RTS + pull-kernel launch + event poll
-    // Same process usually means same device (loopback benches); across
-    // processes assume the conservative single-link xGMI figure.
-    bw = same_proc ? 2.0e12 : 140e9;
+    // Small messages ride the inbox push plane (~20 us one-way measured,
+    // scripts/latency_probe.py); large ones the RTS pull. Bandwidths come
+    // from the cached copy microbenchmark (gpu::calibrate).
+    lat = 20e-6;
+    bw = (same_proc ? gpu::same_gpu_copy_gbps() : gpu::xgmi_link_gbps()) *
+         1e9;
   } else {
     lat = 80e-6;
     bw = 3e9;

@@ -289,6 +289,24 @@ PYBIND11_MODULE(_core, m) {
            py::arg("msg_size"));
 
   m.def("gpu_available", [] { return gpu::available(); });
+  // Run (or re-run) the on-box copy microbenchmark backing evaluate_perf
+  // and persist it to ~/.cache/starway/perf.cal (STARWAY_CALIB_FILE to
+  // override). Returns {same_gpu_gbps, xgmi_gbps} in effect.
+  m.def("calibrate",
+        [](bool force) {
+          std::string err;
+          bool ok;
+          {
+            py::gil_scoped_release rel;
+            ok = gpu::calibrate(force, &err);
+          }
+          if (!ok) throw std::runtime_error("calibrate: " + err);
+          py::dict d;
+          d["same_gpu_gbps"] = gpu::same_gpu_copy_gbps();
+          d["xgmi_gbps"] = gpu::xgmi_link_gbps();
+          return d;
+        },
+        py::arg("force") = false);
   m.def("gpu_device_count", [] { return gpu::device_count(); });
   // IPC hygiene: close every imported hipIpc mapping and drop the handle
   // caches. Call after returning GPU memory to the driver (e.g.

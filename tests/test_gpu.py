@@ -682,3 +682,38 @@ async def test_inbox_cross_process(port):
             p.join()
         p.close()
         await server.aclose()
+
+
+# =============================================================================
+# Calibrated perf model
+# =============================================================================
+
+
+async def test_calibrated_evaluate_perf(tmp_path):
+    import os
+    os.environ["STARWAY_CALIB_FILE"] = str(tmp_path / "perf.cal")
+    vals = _core.calibrate(force=True)
+    assert vals["same_gpu_gbps"] > 100  # HBM-class copy rate
+    assert (tmp_path / "perf.cal").exists()
+    # evaluate_perf must land within 2x of a measured 64 MiB transfer.
+    async with loopback() as (server, client):
+        n = 64 << 20
+        src = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros_like(src)
+        torch.cuda.synchronize()
+        import time as _t
+        # warmup
+        fut = server.arecv(dst, 0, 0)
+        await client.asend(src, 1)
+        await fut
+        t0 = _t.perf_counter()
+        reps = 5
+        for _ in range(reps):
+            fut = server.arecv(dst, 0, 0)
+            await client.asend(src, 1)
+            await fut
+        measured = (_t.perf_counter() - t0) / reps
+        predicted = client.evaluate_perf(n)
+        assert predicted > 0
+        assert measured / 2 <= predicted <= measured * 2, (
+            predicted, measured)
