@@ -124,6 +124,37 @@ def build(verbose: bool = True, debug: bool = False) -> Path:
     return out
 
 
+def build_sanitizer(kind: str = "thread", verbose: bool = True) -> Path:
+    """Sanitizer stress binary: the engine + shm + a GPU stub compiled with
+    g++ -fsanitize={thread,address} and an embedded Python interpreter
+    (csrc/sanitizer_stress_main.cpp). No HIP involved — the engine's
+    cross-thread contracts are identical with or without a device."""
+    BUILD.mkdir(exist_ok=True)
+    (REPO / "bin").mkdir(exist_ok=True)
+    py_inc = sysconfig.get_paths()["include"]
+    ldlib = sysconfig.get_config_var("LDLIBRARY") or ""
+    libdir = sysconfig.get_config_var("LIBDIR") or "/usr/lib"
+    pyver = sysconfig.get_config_var("LDVERSION") or "3.10"
+    out = REPO / "bin" / f"{kind[0]}san_stress"
+    cmd = [
+        "g++", "-O1", "-g", "-std=c++20", f"-fsanitize={kind}",
+        "-fno-omit-frame-pointer",
+        "-I", str(CSRC), "-I", py_inc, "-I", _pybind11_include(),
+        str(CSRC / "engine.cpp"), str(CSRC / "gpu_stub.cpp"),
+        str(CSRC / "sanitizer_stress_main.cpp"),
+        f"-L{libdir}", f"-lpython{pyver}", "-lpthread", "-o", str(out),
+    ]
+    if verbose:
+        print("[build_ext]", " ".join(cmd), flush=True)
+    subprocess.run(cmd, check=True)
+    return out
+
+
 if __name__ == "__main__":
-    build(debug="--debug" in sys.argv)
-    print(f"built {ext_path()}")
+    if "--tsan" in sys.argv:
+        print(f"built {build_sanitizer('thread')}")
+    elif "--asan" in sys.argv:
+        print(f"built {build_sanitizer('address')}")
+    else:
+        build(debug="--debug" in sys.argv)
+        print(f"built {ext_path()}")
