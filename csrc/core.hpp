@@ -582,6 +582,7 @@ class Engine {
     uint64_t tag;
     Op* recv_op;
     uint8_t* dst;  // device dst, or null => pinned bounce
+    int retries = 0;
   };
   std::vector<PendingUnpack> pending_unpacks_;
   struct UnpackBatch {
@@ -599,6 +600,12 @@ class Engine {
     Connection* conn = nullptr;
     uint64_t seq = 0;
   } armed_;
+  // Canceled doorbells whose kernel had not reported by the cancel wait:
+  // polled until they resolve so their pinned cell is never reused while
+  // a late-starting kernel could still write it.
+  std::vector<void*> zombie_arms_;
+  void retire_armed_ticket(void* t);  // free now or park as zombie
+  void poll_zombie_arms();
   // A doorbell copy whose SMSG control frame has not arrived yet (the
   // kernel consumed the message during a disarm race).
   struct ArmedDone {
@@ -727,6 +734,7 @@ void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
 int arm_poll(void* ticket, uint64_t* size_out);
 void arm_cancel(void* ticket);
 void arm_free(void* ticket);
+void arm_leak(void* ticket);  // kernel may still write: do not reuse cell
 // Poll a ticket: 1 done, 0 pending, -1 error.
 int poll_ticket(void* ticket, std::string* err);
 void free_ticket(void* ticket);

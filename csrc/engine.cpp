@@ -763,6 +763,7 @@ void Engine::loop_iteration(bool& did_work) {
   if (!push_batches_.empty()) progress_pushes(did_work);
   if (!unpack_batches_.empty()) progress_unpacks(did_work);
   if (armed_.ticket) progress_armed(did_work);
+  if (!zombie_arms_.empty()) poll_zombie_arms();
   try_arm();
   if (!gpu_pulls_.empty()) poll_gpu(did_work);
   if (!cma_pulls_.empty()) progress_cma(did_work);
@@ -1809,7 +1810,7 @@ void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
       }
     }
     Op* r = armed_.recv_op;
-    gpu::arm_free(armed_.ticket);
+    retire_armed_ticket(armed_.ticket);
     armed_ = Armed{};
     if (st == 1) {
       for (auto it = posted_recvs_.begin(); it != posted_recvs_.end(); ++it)
@@ -1978,10 +1979,19 @@ void Engine::progress_unpacks(bool& did_work) {
       b->remaining--;
       PendingUnpack& m = b->msgs[k];
       if (r < 0) {
-        // Payload never landed (sender died mid-push): the message is
-        // lost; the recv stays pending (unflushed-close contract).
-        repost_recv_front(m.recv_op);
-        inbox_release_seq(m.conn, m.seq);
+        // The in-kernel wait expired before the payload landed. Under a
+        // loaded launch queue the push kernel can start tens of ms late,
+        // so retry (fresh unpack kernel) before declaring the message
+        // lost (sender died mid-push => recv stays pending, the
+        // unflushed-close contract).
+        if (m.retries + 1 < 50 && m.conn && !m.conn->dead) {
+          PendingUnpack again = m;
+          again.retries++;
+          pending_unpacks_.push_back(again);
+        } else {
+          repost_recv_front(m.recv_op);
+          inbox_release_seq(m.conn, m.seq);
+        }
         continue;
       }
       if (!m.dst) {
@@ -2408,6 +2418,28 @@ void Engine::unstage_unexp(UnexpectedMsg* um) {
 // recv. Cancel the wait kernel and see who won. Returns true when the recv
 // is still available; false when the kernel consumed a message into it
 // first (armed_done_ records the copy until its SMSG frame arrives).
+// Free an armed ticket, or park it as a zombie while its kernel could
+// still write the result cell (launch-queue backlog).
+void Engine::retire_armed_ticket(void* t) {
+  uint64_t sz = 0;
+  if (gpu::arm_poll(t, &sz) != 0)
+    gpu::arm_free(t);
+  else
+    zombie_arms_.push_back(t);
+}
+
+void Engine::poll_zombie_arms() {
+  for (size_t i = 0; i < zombie_arms_.size();) {
+    uint64_t sz = 0;
+    if (gpu::arm_poll(zombie_arms_[i], &sz) != 0) {
+      gpu::arm_free(zombie_arms_[i]);
+      zombie_arms_.erase(zombie_arms_.begin() + i);
+    } else {
+      i++;
+    }
+  }
+}
+
 bool Engine::steal_armed(Op* r) {
   gpu::arm_cancel(armed_.ticket);
   uint64_t sz = 0;
@@ -2419,7 +2451,7 @@ bool Engine::steal_armed(Op* r) {
     sched_yield();
   }
   bool copied = (st == 1);
-  gpu::arm_free(armed_.ticket);
+  retire_armed_ticket(armed_.ticket);
   Connection* conn = armed_.conn;
   uint64_t seq = armed_.seq;
   armed_ = Armed{};
@@ -2980,6 +3012,17 @@ void Engine::teardown() {
   if (armed_done_.active) {
     fail_op(armed_done_.recv_op, "operation canceled (endpoint closing)");
     armed_done_ = ArmedDone{};
+  }
+  {
+    auto zdeadline =
+        std::chrono::steady_clock::now() + std::chrono::seconds(2);
+    while (!zombie_arms_.empty() &&
+           std::chrono::steady_clock::now() < zdeadline) {
+      poll_zombie_arms();
+      if (!zombie_arms_.empty()) sched_yield();
+    }
+    for (void* t : zombie_arms_) gpu::arm_leak(t);
+    zombie_arms_.clear();
   }
   for (auto& m : pending_unpacks_)
     fail_op(m.recv_op, "operation canceled (endpoint closing)");

@@ -581,5 +581,10 @@ void arm_free(void* ticket) {
   delete t;
 }
 
+// The wait kernel never started (launch-queue backlog) and may still write
+// its result later: the pinned cell must NOT return to the pool. Leaks one
+// 64 B cell — correctness over thrift.
+void arm_leak(void* ticket) { delete (ArmTicket*)ticket; }
+
 }  // namespace gpu
 }  // namespace sw
