@@ -237,6 +237,31 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
     if lat_samples:
         half_rtt_us = float(np.percentile(np.array(lat_samples) * 1e6, 50)) / 2.0
 
+    # Raw-callback latency variant (N=1 only): the same 64 B tagged
+    # delivery through the public callback API (reference surface too),
+    # with a GIL-yielding wait instead of an asyncio future — isolates the
+    # event-loop wakeup cost from the transport cost.
+    half_rtt_raw_us = None
+    if world == 1:
+        raw_samples = []
+        for i in range(args.lat_iters):
+            done: list = []
+            t1 = time.perf_counter()
+            server.recv(pong, LAT_TAG_BASE + (1 << 28) + i, full_mask,
+                        lambda tag, ln: done.append(1),
+                        lambda err: done.append(err))
+            sent: list = []
+            clients[0].send(ping, LAT_TAG_BASE + (1 << 28) + i,
+                            lambda: sent.append(1),
+                            lambda err: sent.append(err))
+            while not done:
+                time.sleep(0)
+            raw_samples.append(time.perf_counter() - t1)
+            while not sent:
+                time.sleep(0)
+        half_rtt_raw_us = float(
+            np.percentile(np.array(raw_samples) * 1e6, 50)) / 2.0
+
     if world > 1:
         dist.barrier()
 
@@ -245,7 +270,7 @@ async def run_rank(args, rank: int, world: int, device: str, dist):
         await c.aclose()
     await server.aclose()
 
-    return elapsed, half_rtt_us
+    return elapsed, half_rtt_us, half_rtt_raw_us
 
 
 def main() -> int:
@@ -291,7 +316,8 @@ def main() -> int:
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group("gloo")
 
-    elapsed, half_rtt_us = asyncio.run(run_rank(args, rank, world, device, dist))
+    elapsed, half_rtt_us, half_rtt_raw_us = asyncio.run(
+        run_rank(args, rank, world, device, dist))
 
     if rank == 0:
         n_peers = max(1, world - 1)
@@ -324,6 +350,8 @@ def main() -> int:
                 "parallelism": f"p2p-mesh{world}",
                 "pingpong_64B_half_rtt_us":
                     round(half_rtt_us, 2) if half_rtt_us else None,
+                "pingpong_64B_half_rtt_us_raw_api":
+                    round(half_rtt_raw_us, 2) if half_rtt_raw_us else None,
                 "note": "N=1 is same-GPU loopback (HBM-bound); N>=2 is "
                         "xGMI-bound; value counts each sent byte once",
             },
