@@ -503,6 +503,9 @@ class Engine {
   // on: captured from the CONSTRUCTING thread's current device (the
   // engine's own thread never calls hipSetDevice). -1 = no GPU.
   int preferred_device_ = -1;
+  // Per-engine stream lane for the small-message kernels: engines in one
+  // process must not share a doorbell (Wait) stream.
+  int engine_lane_ = 0;
 
  private:
   // Command queue: Python threads -> engine thread.
@@ -710,7 +713,7 @@ struct PushMsg {
 // Batched push into the PEER's inbox; returns an event ticket polled with
 // push_poll / freed with push_free.
 void* inbox_push(const InboxInfo& peer, bool same_proc, int run_device,
-                 const PushMsg* msgs, int n, std::string* err);
+                 const PushMsg* msgs, int n, int lane, std::string* err);
 int push_poll(void* ticket, std::string* err);
 void push_free(void* ticket);
 
@@ -720,7 +723,7 @@ struct UnpackMsg {
   uint8_t* dst;  // device pointer, or null => pinned host bounce
 };
 void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
-                   std::string* err);
+                   int lane, std::string* err);
 int unpack_poll(void* ticket, int idx, std::string* err);  // 0/1/-1
 const uint8_t* unpack_bounce(void* ticket, int idx);
 void unpack_free(void* ticket);
@@ -729,7 +732,7 @@ void unpack_free(void* ticket);
 // into dst on a tag match. arm_poll: 0 running / 1 copied / 2 nomatch /
 // 3 canceled / 4 expired.
 void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
-               uint64_t mask, uint8_t* dst, uint64_t max_size,
+               uint64_t mask, uint8_t* dst, uint64_t max_size, int lane,
                std::string* err);
 int arm_poll(void* ticket, uint64_t* size_out);
 void arm_cancel(void* ticket);

@@ -264,6 +264,8 @@ struct Engine::GpuPull {
 // ---------------------------------------------------------------------------
 
 Engine::Engine(Mode mode) : mode_(mode) {
+  static std::atomic<int> engine_counter{0};
+  engine_lane_ = engine_counter.fetch_add(1, std::memory_order_relaxed);
   if (pipe(wake_fds_) == 0) {
     set_nonblocking(wake_fds_[0]);
     set_nonblocking(wake_fds_[1]);
@@ -1796,7 +1798,7 @@ void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
       // sees the slot within microseconds of the xGMI write landing.
       auto deadline = std::chrono::steady_clock::now() +
                       std::chrono::milliseconds(
-                          (long)env_u64("STARWAY_ARM_WAIT_MS", 5));
+                          (long)env_u64("STARWAY_ARM_WAIT_MS", 2));
       while ((st = gpu::arm_poll(armed_.ticket, &sz)) == 0) {
         if (std::chrono::steady_clock::now() > deadline) break;
       }
@@ -1885,7 +1887,8 @@ void Engine::flush_pending_pushes() {
       }
       std::string err;
       batch->ticket =
-          gpu::inbox_push(c->inbox_r, same_proc, run_dev, msgs, n, &err);
+          gpu::inbox_push(c->inbox_r, same_proc, run_dev, msgs, n,
+                          engine_lane_, &err);
       if (!batch->ticket) {
         // Push plane broken: fail these sends and stop using the inbox.
         c->inbox_r_active = false;
@@ -1952,7 +1955,8 @@ void Engine::flush_pending_unpacks() {
       }
     }
     std::string err;
-    batch->ticket = gpu::inbox_unpack(c->inbox_l, msgs, n, &err);
+    batch->ticket =
+        gpu::inbox_unpack(c->inbox_l, msgs, n, engine_lane_, &err);
     if (!batch->ticket) {
       for (auto& m : batch->msgs) {
         fail_op(m.recv_op, "receive failed: " + err);
@@ -2083,7 +2087,7 @@ void Engine::try_arm() {
   std::string err;
   void* t = gpu::arm_recv(target->inbox_l, target->inbox_seen_seq + 1,
                           r->tag, r->tag_mask, r->buf.ptr, r->buf.size,
-                          &err);
+                          engine_lane_, &err);
   if (!t) {
     SW_DBG("arm failed: %s", err.c_str());
     return;

@@ -55,14 +55,14 @@ bool inbox_create(InboxInfo*, int, std::string* err) {
   return false;
 }
 void inbox_destroy(const InboxInfo&) {}
-void* inbox_push(const InboxInfo&, bool, int, const PushMsg*, int,
+void* inbox_push(const InboxInfo&, bool, int, const PushMsg*, int, int,
                  std::string* err) {
   *err = "no GPU (sanitizer stub)";
   return nullptr;
 }
 int push_poll(void*, std::string*) { return -1; }
 void push_free(void*) {}
-void* inbox_unpack(const InboxInfo&, const UnpackMsg*, int,
+void* inbox_unpack(const InboxInfo&, const UnpackMsg*, int, int,
                    std::string* err) {
   *err = "no GPU (sanitizer stub)";
   return nullptr;
@@ -71,7 +71,7 @@ int unpack_poll(void*, int, std::string*) { return -1; }
 const uint8_t* unpack_bounce(void*, int) { return nullptr; }
 void unpack_free(void*) {}
 void* arm_recv(const InboxInfo&, uint64_t, uint64_t, uint64_t, uint8_t*,
-               uint64_t, std::string* err) {
+               uint64_t, int, std::string* err) {
   *err = "no GPU (sanitizer stub)";
   return nullptr;
 }

@@ -40,6 +40,7 @@
 #include <algorithm>
 #include <map>
 #include <mutex>
+#include <tuple>
 #include <vector>
 
 namespace sw {
@@ -225,9 +226,15 @@ __global__ __launch_bounds__(256) void k_inbox_wait(ArmArgs a) {
 
 enum class SmStream { Push, Unpack, Wait };
 
-static hipStream_t sm_stream(int device, SmStream kind) {
-  static std::map<std::pair<int, int>, hipStream_t> streams;  // sm_mu held
-  auto key = std::make_pair(device, (int)kind);
+// Streams are keyed by (device, kind, lane). The lane is the calling
+// engine's identity (mod 8): two engines in one process (loopback pairs)
+// must NEVER share a Wait stream — a doorbell kernel spins for up to its
+// bound, and a second engine's doorbell queued behind it on the same
+// stream cannot start, serializing bidirectional latency paths into the
+// expiry cadence (measured: pingpong-flag RTT 116 us median, 5 ms tail).
+static hipStream_t sm_stream(int device, SmStream kind, int lane) {
+  static std::map<std::tuple<int, int, int>, hipStream_t> streams;  // sm_mu
+  auto key = std::make_tuple(device, (int)kind, lane & 7);
   auto it = streams.find(key);
   if (it != streams.end()) return it->second;
   int prev;
@@ -364,7 +371,7 @@ struct PushTicket {
 };
 
 void* inbox_push(const InboxInfo& peer, bool same_proc, int run_device,
-                 const PushMsg* msgs, int n, std::string* err) {
+                 const PushMsg* msgs, int n, int lane, std::string* err) {
   std::lock_guard<std::mutex> lk(sm_mu);
   if (n < 1 || n > kPushMax) {
     *err = "inbox_push: bad batch size";
@@ -384,7 +391,7 @@ void* inbox_push(const InboxInfo& peer, bool same_proc, int run_device,
   int prev;
   hipGetDevice(&prev);
   hipSetDevice(run_device);
-  hipStream_t stream = sm_stream(run_device, SmStream::Push);
+  hipStream_t stream = sm_stream(run_device, SmStream::Push, lane);
   hipLaunchKernelGGL(k_inbox_push, dim3(n), dim3(256), 0, stream, args);
   hipError_t e = hipGetLastError();
   PushTicket* t = nullptr;
@@ -435,7 +442,7 @@ struct UnpackTicket {
 };
 
 void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
-                   std::string* err) {
+                   int lane, std::string* err) {
   std::lock_guard<std::mutex> lk(sm_mu);
   init_pools();
   if (n < 1 || n > kPushMax) {
@@ -470,7 +477,7 @@ void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
   int prev;
   hipGetDevice(&prev);
   hipSetDevice(mine.device);
-  hipStream_t stream = sm_stream(mine.device, SmStream::Unpack);
+  hipStream_t stream = sm_stream(mine.device, SmStream::Unpack, lane);
   hipLaunchKernelGGL(k_inbox_unpack, dim3(n), dim3(256), 0, stream, args);
   hipError_t e = hipGetLastError();
   hipSetDevice(prev);
@@ -519,7 +526,7 @@ struct ArmTicket {
 };
 
 void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
-               uint64_t mask, uint8_t* dst, uint64_t max_size,
+               uint64_t mask, uint8_t* dst, uint64_t max_size, int lane,
                std::string* err) {
   std::lock_guard<std::mutex> lk(sm_mu);
   init_pools();
@@ -548,7 +555,7 @@ void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
   int prev;
   hipGetDevice(&prev);
   hipSetDevice(mine.device);
-  hipStream_t stream = sm_stream(mine.device, SmStream::Wait);
+  hipStream_t stream = sm_stream(mine.device, SmStream::Wait, lane);
   hipLaunchKernelGGL(k_inbox_wait, dim3(1), dim3(256), 0, stream, a);
   hipError_t e = hipGetLastError();
   hipSetDevice(prev);
