@@ -1809,6 +1809,12 @@ void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
         while ((st = gpu::arm_poll(armed_.ticket, &sz)) == 0 &&
                std::chrono::steady_clock::now() < hard)
           sched_yield();
+        // Diagnostic (STARWAY_DEBUG_ARM=1): a doorbell that had to be
+        // canceled because it never reported, with its final state.
+        static const bool dbg_arm = getenv("STARWAY_DEBUG_ARM") != nullptr;
+        if (dbg_arm)
+          fprintf(stderr, "[sw-arm] seq=%llu missed, final st=%d\n",
+                  (unsigned long long)seq, st);
       }
     }
     Op* r = armed_.recv_op;
