@@ -1791,8 +1791,11 @@ void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
   }
   // Doorbell fast path: a pre-armed kernel is watching exactly this seq.
   if (armed_.ticket && armed_.conn == c && armed_.seq == seq) {
+    static const bool dbg_arm2 = getenv("STARWAY_DEBUG_ARM") != nullptr;
+    auto dbg_t0 = std::chrono::steady_clock::now();
     uint64_t sz = 0;
     int st = gpu::arm_poll(armed_.ticket, &sz);
+    int st_first = st;
     if (st == 0) {
       // The payload and this control frame race each other; the kernel
       // sees the slot within microseconds of the xGMI write landing.
@@ -1816,6 +1819,15 @@ void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
           fprintf(stderr, "[sw-arm] seq=%llu missed, final st=%d\n",
                   (unsigned long long)seq, st);
       }
+    }
+    if (dbg_arm2) {
+      double us = std::chrono::duration<double, std::micro>(
+                      std::chrono::steady_clock::now() - dbg_t0)
+                      .count();
+      if (us > 100 || st != 1)
+        fprintf(stderr,
+                "[sw-arm] seq=%llu first_st=%d final_st=%d wait=%.0fus\n",
+                (unsigned long long)seq, st_first, st, us);
     }
     Op* r = armed_.recv_op;
     retire_armed_ticket(armed_.ticket);

@@ -15,6 +15,7 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 import numpy as np  # noqa: E402
 
 PING, PONG = 0x2B20, 0x2B21
+MODE = "worker"
 FULL = (1 << 64) - 1
 
 
@@ -33,8 +34,12 @@ async def run(device, iters, nbytes):
 
     server = sw.Server()
     client = sw.Client()
-    addr = server.listen_address()
-    await client.aconnect_address(addr)
+    if MODE == "socket":
+        server.listen("0.0.0.0", 18987)
+        await client.aconnect("127.0.0.1", 18987)
+    else:
+        addr = server.listen_address()
+        await client.aconnect_address(addr)
     ep = next(iter(server.list_clients()))
 
     ping = alloc(device, nbytes, 1)
@@ -48,14 +53,20 @@ async def run(device, iters, nbytes):
             await server.asend(ep, pong_b, PONG)
 
     samples = []
+    send_us = []
+    reply_us = []
 
     async def client_side():
         for _ in range(iters):
             reply = client.arecv(rx_c, PONG, FULL)
             t0 = time.perf_counter()
             await client.asend(ping, PING)
+            t1 = time.perf_counter()
             await reply
-            samples.append(time.perf_counter() - t0)
+            t2 = time.perf_counter()
+            samples.append(t2 - t0)
+            send_us.append((t1 - t0) * 1e6)
+            reply_us.append((t2 - t1) * 1e6)
 
     await asyncio.gather(server_side(), client_side())
     ss = server._server.get_stats()
@@ -69,6 +80,8 @@ async def run(device, iters, nbytes):
         "p50_us": round(float(np.percentile(us, 50)), 1),
         "p90_us": round(float(np.percentile(us, 90)), 1),
         "max_us": round(float(us.max()), 1),
+        "send_p50_us": round(float(np.percentile(send_us, 50)), 1),
+        "reply_p50_us": round(float(np.percentile(reply_us, 50)), 1),
         "server": {k: v for k, v in ss.items() if v},
         "client": {k: v for k, v in cs.items() if v},
     }))
@@ -80,5 +93,7 @@ if __name__ == "__main__":
     p.add_argument("--device", default="cpu")
     p.add_argument("--iters", type=int, default=300)
     p.add_argument("--nbytes", type=int, default=1)
+    p.add_argument("--mode", default="worker", choices=("worker", "socket"))
     args = p.parse_args()
+    globals()["MODE"] = args.mode
     asyncio.run(run(args.device, args.iters, args.nbytes))
