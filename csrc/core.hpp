@@ -603,6 +603,12 @@ class Engine {
     Connection* conn = nullptr;
     uint64_t seq = 0;
   } armed_;
+  // Doorbell re-arm backoff: after a few consecutive unused expiries the
+  // engine stops relaunching until inbox traffic resumes or a recv is
+  // posted — back-to-back spinning doorbells monopolize their hardware
+  // queue and starve co-mapped streams under queue oversubscription.
+  int arm_streak_ = 0;
+  bool arm_backoff_ = false;
   // Canceled doorbells whose kernel had not reported by the cancel wait:
   // polled until they resolve so their pinned cell is never reused while
   // a late-starting kernel could still write it.

@@ -1771,6 +1771,8 @@ void Engine::dispatch_smsg_to_recv(Connection* c, Op* r, uint64_t tag,
 void Engine::handle_smsg(Connection* c, uint64_t tag, uint64_t size,
                          uint64_t seq) {
   c->inbox_seen_seq = seq;
+  arm_streak_ = 0;
+  arm_backoff_ = false;
   // Doorbell copy that raced a disarm: this frame carries its tag.
   if (armed_done_.active && armed_done_.conn == c && armed_done_.seq == seq) {
     Op* r = armed_done_.recv_op;
@@ -2076,7 +2078,8 @@ void Engine::try_arm() {
     const char* v = getenv("STARWAY_DOORBELL");
     return !(v && !strcmp(v, "0"));
   }();
-  if (!arm_on || armed_.ticket || armed_done_.active) return;
+  if (!arm_on || armed_.ticket || armed_done_.active || arm_backoff_)
+    return;
   // Latency pattern only: with several recvs outstanding (throughput
   // pattern) a doorbell would serialize the batched unpack path into one
   // arm launch + inline wait per message.
@@ -2118,10 +2121,18 @@ void Engine::progress_armed(bool& did_work) {
   uint64_t sz = 0;
   int st = gpu::arm_poll(armed_.ticket, &sz);
   if (st == 4) {
-    // Bounded wait expired with no message: re-arm at the same sequence.
+    // Bounded wait expired with no message: re-arm at the same sequence,
+    // but only a few times in a row — continuous re-arming keeps a
+    // spinning kernel resident ~100% of the time, which starves streams
+    // sharing its hardware queue. After the streak the doorbell stands
+    // down until traffic resumes (handle_smsg) or a recv is posted.
     did_work = true;
     gpu::arm_free(armed_.ticket);
     armed_ = Armed{};
+    if (++arm_streak_ >= 3) {
+      arm_backoff_ = true;
+      return;
+    }
     try_arm();
   }
   // 1 (copied) / 2 (nomatch) are resolved by handle_smsg when the control
@@ -2529,6 +2540,8 @@ void Engine::repost_recv_front(Op* r) {
 }
 
 void Engine::match_or_stash_recv(Op* op) {
+  arm_streak_ = 0;
+  arm_backoff_ = false;  // fresh recv: the doorbell may re-engage
   if (!try_match_unexpected(op)) posted_recvs_.push_back(op);
 }
 

@@ -36,6 +36,15 @@ from typing import Any, Literal
 # shares one copy. STARWAY_USE_SYSTEM_HIP=true opts into the system ROCm
 # runtime instead (only safe in torch-free processes).
 # ---------------------------------------------------------------------------
+# The engine uses several concurrent streams per device (pull lanes +
+# small-message push/unpack/doorbell lanes). ROCm multiplexes streams onto
+# GPU_MAX_HW_QUEUES hardware queues (default 4); once oversubscribed,
+# co-mapped streams are time-sliced at ~ms granularity and a resident
+# doorbell kernel can starve a push stream (measured: bistable 3.3 ms
+# pingpong RTT). Raise the default before the HIP runtime initializes;
+# a user-set value is respected.
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
+
 _used_hip = "system"
 if os.environ.get("STARWAY_USE_SYSTEM_HIP", "false") != "true":
     _torch_spec = find_spec("torch")
