@@ -81,3 +81,9 @@ void arm_free(void*) {}
 
 }  // namespace gpu
 }  // namespace sw
+
+namespace sw {
+namespace gpu {
+void arm_leak(void*) {}
+}  // namespace gpu
+}  // namespace sw
