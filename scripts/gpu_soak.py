@@ -55,6 +55,9 @@ async def main(seconds: int) -> None:
         if it % 500 == 0:
             torch.cuda.synchronize()
             assert torch.equal(src, dst), f"mismatch at iter {it} size {n}"
+        if it % 100_000 == 0:
+            rss = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+            print(f"  it={it} rss={(rss - rss0) / 1024:.1f} MB", flush=True)
 
     torch.cuda.synchronize()
     free1, _ = torch.cuda.mem_get_info()
