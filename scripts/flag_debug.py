@@ -4,6 +4,14 @@ the scenario exactly (same tag each iteration, bidirectional pre-posted
 recvs, one event loop driving both sides)."""
 from __future__ import annotations
 
+import os
+
+# ROCm multiplexes streams onto GPU_MAX_HW_QUEUES hardware queues
+# (default 4); oversubscription time-slices co-mapped streams at ~ms
+# granularity. Must be set before the FIRST HIP init in the process
+# (torch's or ours) — see ROUND2_NOTES.md "hardware-queue starvation".
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
+
 import asyncio
 import json
 import sys

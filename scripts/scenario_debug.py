@@ -4,6 +4,14 @@ gather of N concurrent sends/recvs per batch) with per-batch timing and
 engine stats, to localize where batch time goes."""
 from __future__ import annotations
 
+import os
+
+# ROCm multiplexes streams onto GPU_MAX_HW_QUEUES hardware queues
+# (default 4); oversubscription time-slices co-mapped streams at ~ms
+# granularity. Must be set before the FIRST HIP init in the process
+# (torch's or ours) — see ROUND2_NOTES.md "hardware-queue starvation".
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
+
 import argparse
 import asyncio
 import gc

@@ -14,6 +14,14 @@ Writes a JSON + text table to --out (default gpurun_out/sweep.json).
 """
 from __future__ import annotations
 
+import os
+
+# ROCm multiplexes streams onto GPU_MAX_HW_QUEUES hardware queues
+# (default 4); oversubscription time-slices co-mapped streams at ~ms
+# granularity. Must be set before the FIRST HIP init in the process
+# (torch's or ours) — see ROUND2_NOTES.md "hardware-queue starvation".
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
+
 import argparse
 import asyncio
 import json
