@@ -1,11 +1,3 @@
-import os
-
-# ROCm multiplexes streams onto GPU_MAX_HW_QUEUES hardware queues
-# (default 4); oversubscription time-slices co-mapped streams at ~ms
-# granularity. Must be set before the FIRST HIP init in the process
-# (torch's or ours) — see ROUND2_NOTES.md "hardware-queue starvation".
-os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
-
 """Test harness config.
 
 * registers the ``gpu`` marker (tests needing an MI355X; skipped on hosts
@@ -15,6 +7,14 @@ os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
   a per-test timeout.
 """
 from __future__ import annotations
+
+import os
+
+# ROCm multiplexes streams onto GPU_MAX_HW_QUEUES hardware queues
+# (default 4); oversubscription time-slices co-mapped streams at ~ms
+# granularity. Must be set before the FIRST HIP init in the process
+# (torch's or ours) — see ROUND2_NOTES.md "hardware-queue starvation".
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "16")
 
 import asyncio
 import inspect
