@@ -717,3 +717,95 @@ async def test_calibrated_evaluate_perf(tmp_path):
         assert predicted > 0
         assert measured / 2 <= predicted <= measured * 2, (
             predicted, measured)
+
+
+def _inbox_flush_close_sender(port: int):
+    import torch
+
+    import starway_amd as sw
+
+    async def inner():
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+        for i in range(8):
+            src = torch.full((512,), 40 + i, dtype=torch.uint8,
+                             device="cuda")
+            torch.cuda.synchronize()
+            await client.asend(src, 70 + i)
+        await client.aflush()
+        await client.aclose()  # sender gone before any recv is posted
+
+    asyncio.run(inner())
+
+
+async def test_inbox_flush_then_close_delivers(port):
+    # Flush guarantees delivery even if the sender closes before the
+    # receiver posts recvs: the payloads already live in the receiver's
+    # ring and the control frames are drained before the connection dies.
+    server = sw.Server()
+    server.listen("127.0.0.1", port)
+    ctx = mp.get_context("spawn")
+    p = ctx.Process(target=_inbox_flush_close_sender, args=(port,))
+    p.start()
+    try:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+        await asyncio.sleep(0.2)  # conn death observed; slots must survive
+        for i in range(8):
+            dst = torch.zeros(512, dtype=torch.uint8, device="cuda")
+            torch.cuda.synchronize()
+            tag, ln = await asyncio.wait_for(
+                server.arecv(dst, 70 + i, (1 << 64) - 1), timeout=30)
+            torch.cuda.synchronize()
+            assert ln == 512 and torch.all(dst == 40 + i)
+    finally:
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
+        await server.aclose()
+
+
+def _no_inbox_child_client(port: int):
+    import os
+
+    os.environ["STARWAY_INBOX"] = "0"
+    import torch
+
+    import starway_amd as sw
+
+    async def inner():
+        client = sw.Client()
+        await client.aconnect("127.0.0.1", port)
+        src = torch.full((1024,), 55, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        await client.asend(src, 91)  # RTS path: completes at delivery
+        await client.aflush()
+        await client.aclose()
+
+    asyncio.run(inner())
+
+
+async def test_small_device_send_with_inbox_disabled(port):
+    # STARWAY_INBOX=0 on either side falls back to the RTS pull plane.
+    server = sw.Server()
+    server.listen("127.0.0.1", port)
+    ctx = mp.get_context("spawn")
+    p = ctx.Process(target=_no_inbox_child_client, args=(port,))
+    p.start()
+    try:
+        dst = torch.zeros(1024, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        tag, ln = await asyncio.wait_for(
+            server.arecv(dst, 91, (1 << 64) - 1), timeout=60)
+        torch.cuda.synchronize()
+        assert ln == 1024 and torch.all(dst == 55)
+        # The message rode the rendezvous plane, not the ring.
+        assert server._server.get_stats()["gpu_rx"] >= 1
+    finally:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
+        await server.aclose()
