@@ -966,8 +966,12 @@ void Engine::start_send(Op* op, Connection* c) {
   // FT_SMSG control frame is enqueued HERE so per-connection message order
   // is preserved against eager/RTS traffic. The op completes when the push
   // kernel's event lands (payload captured out of the user buffer).
-  if (op->buf.device >= 0 && op->buf.rows == 0 && op->buf.size > 0 &&
-      !force_xhost && c->inbox_r_active &&
+  static const bool inbox_tx_on = [] {
+    const char* v = getenv("STARWAY_INBOX");
+    return !(v && !strcmp(v, "0"));
+  }();
+  if (inbox_tx_on && op->buf.device >= 0 && op->buf.rows == 0 &&
+      op->buf.size > 0 && !force_xhost && c->inbox_r_active &&
       op->buf.size <= c->inbox_r.slot_bytes - gpu::kInboxHdrBytes &&
       c->inbox_next_seq <= c->inbox_credit_base + c->inbox_r.slots) {
     uint64_t seq = c->inbox_next_seq++;
