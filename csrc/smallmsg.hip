@@ -470,7 +470,17 @@ void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
     args.d[i].seq = msgs[i].seq;
     args.d[i].size = msgs[i].size;
     uint8_t* dst = msgs[i].dst;
-    if (!dst) dst = g_bounce_pool.get();  // host recv: pinned staging
+    if (!dst) {
+      dst = g_bounce_pool.get();  // host recv: pinned staging
+      if (!dst) {
+        g_bounce_pool.put((uint8_t*)t->results);
+        for (auto* b : t->bounces)
+          if (b) g_bounce_pool.put(b);
+        delete t;
+        *err = "pinned pool exhausted";
+        return nullptr;
+      }
+    }
     args.d[i].dst = dst;
     t->bounces.push_back(msgs[i].dst ? nullptr : dst);
   }
