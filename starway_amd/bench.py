@@ -1,17 +1,25 @@
 """Benchmark CLI — ``python -m starway_amd.bench``.
 
-Role/orchestration parity with the reference CLI (reference
-src/starway/bench.py: roles server|client|loopback, a JSON-over-tagged-
-messages control protocol with READY/DONE handshakes, per-scenario
-overrides, JSON reports). Differences: ``--device cpu|cuda`` selects host
-numpy buffers vs HIP device tensors (the reference had no GPU path), and
-``--tls`` is gone — the transport is always our native TCP+xGMI stack.
+Role/orchestration parity with the reference CLI: roles
+``server|client|loopback``, a JSON-over-tagged-messages control protocol
+with READY/DONE handshakes, per-scenario overrides, JSON reports.
+Differences from the reference: ``--device cpu|cuda`` selects host numpy
+buffers vs HIP device tensors (the reference had no GPU path), and
+``--tls`` is accepted but inert — the transport is always the native
+TCP + shm-ring + CMA + xGMI stack.
+
+Wire protocol (one control channel per connected pair):
+  client --CONTROL_TAG--> {"scenario": name, "config": {...}}   per run
+  server --READY_TAG----> 1 byte        (scenario server is listening)
+  <both run the scenario's data traffic on its own tags>
+  server --DONE_TAG-----> 1 byte        (server-side runner finished)
+  client --CONTROL_TAG--> {"scenario": "__shutdown__"}          at the end
 """
 from __future__ import annotations
 
 import argparse
-import gc
 import asyncio
+import gc
 import json
 import os
 import sys
@@ -32,23 +40,27 @@ from .benchmarks.scenarios import (
     ScenarioResult,
 )
 
+SHUTDOWN = "__shutdown__"
+
+
+# ---------------------------------------------------------------------------
+# CLI
+# ---------------------------------------------------------------------------
 
 def parse_size(value: str) -> int:
     text = value.strip().lower().replace("_", "")
-    suffixes = {
-        "k": 1024, "kb": 1024, "ki": 1024, "kib": 1024,
-        "m": 1024 ** 2, "mb": 1024 ** 2, "mi": 1024 ** 2, "mib": 1024 ** 2,
-        "g": 1024 ** 3, "gb": 1024 ** 3, "gi": 1024 ** 3, "gib": 1024 ** 3,
-    }
-    for suffix, mult in suffixes.items():
+    for suffix, mult in (("kib", 1024), ("kb", 1024), ("ki", 1024),
+                         ("k", 1024), ("mib", 1024 ** 2), ("mb", 1024 ** 2),
+                         ("mi", 1024 ** 2), ("m", 1024 ** 2),
+                         ("gib", 1024 ** 3), ("gb", 1024 ** 3),
+                         ("gi", 1024 ** 3), ("g", 1024 ** 3)):
         if text.endswith(suffix):
             return int(float(text[: -len(suffix)]) * mult)
     return int(float(text))
 
 
 def parse_worker_address(value: str) -> bytes:
-    cleaned = value.replace(":", "").replace(" ", "").strip()
-    return bytes.fromhex(cleaned)
+    return bytes.fromhex(value.replace(":", "").replace(" ", "").strip())
 
 
 def build_parser() -> argparse.ArgumentParser:
@@ -86,163 +98,162 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--flag-iterations", type=int)
     p.add_argument("--flag-warmup", type=int)
     p.add_argument("--stream-bytes", type=parse_size)
-    p.add_argument("--stream-iterations", type=int)
     p.add_argument("--stream-warmup", type=int)
+    p.add_argument("--stream-iterations", type=int)
     p.add_argument("--output", type=Path)
     p.add_argument("--store-trace", action="store_true")
     return p
 
 
+_OVERRIDE_KNOBS = {
+    "large-array": {"message_bytes": "large_bytes",
+                    "iterations": "large_iterations",
+                    "warmup": "large_warmup"},
+    "small-messages": {"message_bytes": "small_bytes",
+                       "iterations": "small_iterations",
+                       "warmup_batches": "small_warmup",
+                       "concurrency": "small_concurrency"},
+    "pingpong-flag": {"iterations": "flag_iterations",
+                      "warmup": "flag_warmup"},
+    "streaming-duplex": {"message_bytes": "stream_bytes",
+                         "iterations": "stream_iterations",
+                         "warmup": "stream_warmup"},
+}
+
+
 def scenario_plan(args: argparse.Namespace) -> list[tuple[str, dict[str, Any]]]:
+    wanted: Sequence[str]
     if not args.scenarios or (
         len(args.scenarios) == 1 and args.scenarios[0].lower() == "all"
     ):
-        requested: Sequence[str] = list_scenarios()
+        wanted = list_scenarios()
     else:
-        requested = args.scenarios
-
-    plan: list[tuple[str, dict[str, Any]]] = []
-    for name in requested:
+        wanted = args.scenarios
+    plan = []
+    for name in wanted:
         if name not in SCENARIOS:
             raise ValueError(
                 f"Unknown scenario '{name}'. Available: {', '.join(list_scenarios())}"
             )
-        ov: dict[str, Any] = {}
-        if name == "large-array":
-            ov = {"message_bytes": args.large_bytes,
-                  "iterations": args.large_iterations,
-                  "warmup": args.large_warmup}
-        elif name == "small-messages":
-            ov = {"message_bytes": args.small_bytes,
-                  "iterations": args.small_iterations,
-                  "warmup_batches": args.small_warmup,
-                  "concurrency": args.small_concurrency}
-        elif name == "pingpong-flag":
-            ov = {"iterations": args.flag_iterations,
-                  "warmup": args.flag_warmup}
-        elif name == "streaming-duplex":
-            ov = {"message_bytes": args.stream_bytes,
-                  "iterations": args.stream_iterations,
-                  "warmup": args.stream_warmup}
-        ov = {k: v for k, v in ov.items() if v is not None}
+        overrides = {
+            knob: getattr(args, attr)
+            for knob, attr in _OVERRIDE_KNOBS.get(name, {}).items()
+            if getattr(args, attr) is not None
+        }
         if args.device:
-            ov["device"] = args.device
-        plan.append((name, ov))
+            overrides["device"] = args.device
+        plan.append((name, overrides))
     return plan
 
 
-def encode_control(payload: Mapping[str, Any]) -> np.ndarray:
-    data = json.dumps(payload, separators=(",", ":"), sort_keys=True).encode()
-    return np.frombuffer(data, dtype=np.uint8).copy()
+# ---------------------------------------------------------------------------
+# Control plane: JSON frames + READY/DONE flags over reserved tags
+# ---------------------------------------------------------------------------
+
+def _to_frame(payload: Mapping[str, Any]) -> np.ndarray:
+    blob = json.dumps(payload, separators=(",", ":"), sort_keys=True).encode()
+    return np.frombuffer(blob, dtype=np.uint8).copy()
 
 
-def decode_control(buffer: np.ndarray, length: int) -> Mapping[str, Any]:
-    return json.loads(memoryview(buffer)[:length].tobytes().decode())
+class ClientPeer:
+    """Client half of a bench session: drives scenarios and the control
+    protocol. Doubles as the scenarios' ClientRuntime (attributes
+    ``client`` / ``tag_mask`` + ``flush``)."""
 
-
-class ClientSession:
-    def __init__(self, client: Client):
+    def __init__(self, client: Client) -> None:
         self.client = client
         self.tag_mask = TAG_MASK
-        self._ready = np.zeros(1, dtype=np.uint8)
-        self._done = np.zeros(1, dtype=np.uint8)
-
-    async def send_control(self, payload: Mapping[str, Any]) -> None:
-        await self.client.asend(encode_control(payload), CONTROL_TAG)
-        await self.flush()
-
-    async def wait_ready(self) -> None:
-        await self.client.arecv(self._ready, READY_TAG, self.tag_mask)
-
-    async def wait_done(self) -> None:
-        await self.client.arecv(self._done, DONE_TAG, self.tag_mask)
+        self._flag = np.zeros(1, dtype=np.uint8)
 
     async def flush(self) -> None:
         await self.client.aflush()
 
+    async def request(self, payload: Mapping[str, Any]) -> None:
+        await self.client.asend(_to_frame(payload), CONTROL_TAG)
+        await self.client.aflush()
 
-class ClientScenarioContext:
-    def __init__(self, session: ClientSession):
-        self._session = session
-        self.client = session.client
-        self.tag_mask = session.tag_mask
+    async def await_flag(self, tag: int) -> None:
+        await self.client.arecv(self._flag, tag, self.tag_mask)
 
-    async def flush(self) -> None:
-        await self._session.flush()
+    async def run_plan(self, plan) -> list[ScenarioResult]:
+        results = []
+        for name, overrides in plan:
+            print(f"[client] Starting '{name}' with {overrides or 'defaults'}.")
+            await self.request({"scenario": name, "config": overrides})
+            await self.await_flag(READY_TAG)
+            # Bench hygiene: a gen-2 GC pause (~40 ms from future/closure
+            # churn) would dominate a batch sample; collect up front and
+            # keep the collector off inside the timed region.
+            gc.collect()
+            gc.disable()
+            try:
+                results.append(
+                    await get_scenario(name).client_runner(self, overrides))
+            finally:
+                gc.enable()
+            await self.await_flag(DONE_TAG)
+            print(f"[client] Completed '{name}'.")
+        await self.request({"scenario": SHUTDOWN})
+        return results
 
 
-class ServerSession:
-    def __init__(self, server: Server, endpoint):
+class ServerPeer:
+    """Server half: executes scenario requests until shutdown. Doubles as
+    the scenarios' ServerRuntime (``server`` / ``endpoint`` / ``tag_mask``
+    + ``signal_ready`` / ``flush_endpoint``)."""
+
+    def __init__(self, server: Server, endpoint) -> None:
         self.server = server
         self.endpoint = endpoint
         self.tag_mask = TAG_MASK
-        self._ready = np.array([1], dtype=np.uint8)
-        self._done = np.array([1], dtype=np.uint8)
-
-    async def recv_control(self, max_bytes: int = 4096) -> Mapping[str, Any]:
-        buffer = np.empty(max_bytes, dtype=np.uint8)
-        _, length = await self.server.arecv(buffer, CONTROL_TAG, self.tag_mask)
-        return decode_control(buffer, length)
-
-    async def send_ready(self) -> None:
-        await self.server.asend(self.endpoint, self._ready, READY_TAG)
-
-    async def send_done(self) -> None:
-        await self.server.asend(self.endpoint, self._done, DONE_TAG)
-
-
-class ServerScenarioContext:
-    def __init__(self, session: ServerSession):
-        self._session = session
-        self.server = session.server
-        self.endpoint = session.endpoint
-        self.tag_mask = session.tag_mask
+        self._flag = np.ones(1, dtype=np.uint8)
 
     async def signal_ready(self) -> None:
-        await self._session.send_ready()
+        await self.server.asend(self.endpoint, self._flag, READY_TAG)
 
     async def flush_endpoint(self) -> None:
         await self.server.aflush_ep(self.endpoint)
 
+    async def next_request(self, max_bytes: int = 4096) -> Mapping[str, Any]:
+        sink = np.empty(max_bytes, dtype=np.uint8)
+        _, n = await self.server.arecv(sink, CONTROL_TAG, self.tag_mask)
+        return json.loads(sink[:n].tobytes().decode())
+
+    async def serve_until_shutdown(self) -> None:
+        while True:
+            request = await self.next_request()
+            name = request.get("scenario")
+            if name == SHUTDOWN:
+                print("[server] Shutdown request received.")
+                return
+            if name not in SCENARIOS:
+                raise ValueError(f"Unknown scenario '{name}' from client.")
+            overrides = request.get("config", {})
+            print(f"[server] Running '{name}' with {overrides or 'defaults'}.")
+            await get_scenario(name).server_runner(self, overrides)
+            await self.server.asend(self.endpoint, self._flag, DONE_TAG)
+            print(f"[server] Scenario '{name}' completed.")
+
+
+# ---------------------------------------------------------------------------
+# Roles
+# ---------------------------------------------------------------------------
 
 async def run_client(args: argparse.Namespace) -> list[ScenarioResult]:
     client = Client()
-    results: list[ScenarioResult] = []
     try:
         if args.connect_mode == "worker":
             if not args.worker_address:
                 raise ValueError("--worker-address required for connect-mode=worker")
-            addr = parse_worker_address(args.worker_address)
-            await client.aconnect_address(addr)
-            print(f"[client] Connected via worker address ({len(addr)} bytes).")
+            blob = parse_worker_address(args.worker_address)
+            await client.aconnect_address(blob)
+            print(f"[client] Connected via worker address ({len(blob)} bytes).")
         else:
             await client.aconnect(args.server_host, args.port)
             print(f"[client] Connected to {args.server_host}:{args.port}.")
-
-        session = ClientSession(client)
-        context = ClientScenarioContext(session)
-
-        for name, overrides in scenario_plan(args):
-            scenario = get_scenario(name)
-            print(f"[client] Starting '{name}' with {overrides or 'defaults'}.")
-            await session.send_control({"scenario": name, "config": overrides})
-            await session.wait_ready()
-            # Bench hygiene: a gen-2 GC pause (~40 ms, triggered by the
-            # future/closure churn of concurrent-op scenarios) would
-            # dominate a batch sample; collect up front, disable during
-            # the timed region.
-            gc.collect()
-            gc.disable()
-            try:
-                result = await scenario.client_runner(context, overrides)
-            finally:
-                gc.enable()
-            results.append(result)
-            await session.wait_done()
-            print(f"[client] Completed '{name}'.")
-
-        await session.send_control({"scenario": "__shutdown__"})
-        await session.flush()
+        peer = ClientPeer(client)
+        results = await peer.run_plan(scenario_plan(args))
+        await peer.flush()
     finally:
         await client.aclose()
     return results
@@ -251,83 +262,60 @@ async def run_client(args: argparse.Namespace) -> list[ScenarioResult]:
 async def run_server(args: argparse.Namespace) -> None:
     server = Server()
     loop = asyncio.get_running_loop()
-    accepted: asyncio.Queue = asyncio.Queue()
+    first_ep: asyncio.Future = loop.create_future()
     server.set_accept_cb(
-        lambda ep: loop.call_soon_threadsafe(accepted.put_nowait, ep)
-    )
+        lambda ep: loop.call_soon_threadsafe(
+            lambda: first_ep.done() or first_ep.set_result(ep)))
 
     if args.listen_mode == "worker":
-        worker_address = server.listen_address()
-        print(f"[server] Listening via worker address: {worker_address.hex()}")
+        blob = server.listen_address()
+        print(f"[server] Listening via worker address: {blob.hex()}")
     else:
         server.listen(args.addr, args.port)
         print(f"[server] Listening on {args.addr}:{args.port}")
 
-    endpoint = await accepted.get()
+    endpoint = await first_ep
     print("[server] Client accepted.")
-    session = ServerSession(server, endpoint)
     try:
-        while True:
-            control = await session.recv_control()
-            name = control.get("scenario")
-            if name == "__shutdown__":
-                print("[server] Shutdown request received.")
-                break
-            if name not in SCENARIOS:
-                raise ValueError(f"Unknown scenario '{name}' from client.")
-            overrides = control.get("config", {})
-            scenario = get_scenario(name)
-            print(f"[server] Running '{name}' with {overrides or 'defaults'}.")
-            await scenario.server_runner(ServerScenarioContext(session), overrides)
-            await session.send_done()
-            print(f"[server] Scenario '{name}' completed.")
+        await ServerPeer(server, endpoint).serve_until_shutdown()
     finally:
         await server.aclose()
         print("[server] Closed.")
 
 
 async def run_loopback(args: argparse.Namespace) -> list[ScenarioResult]:
-    client_done: asyncio.Future = asyncio.get_running_loop().create_future()
-
-    async def client_task() -> None:
-        try:
-            client_done.set_result(await run_client(args))
-        except Exception as exc:
-            if not client_done.done():
-                client_done.set_exception(exc)
-            raise
-
-    server_task = asyncio.create_task(run_server(args))
-    client_fut = asyncio.create_task(client_task())
+    # Both halves share one event loop and talk over localhost.
+    server_side = asyncio.create_task(run_server(args))
     try:
-        results = await client_done
+        return await run_client(args)
     finally:
-        await client_fut
-        await server_task
-    return results
+        await server_side
 
 
-def dump_results(results: Sequence[ScenarioResult], args: argparse.Namespace) -> None:
+# ---------------------------------------------------------------------------
+# Reporting
+# ---------------------------------------------------------------------------
+
+def dump_results(results: Sequence[ScenarioResult],
+                 args: argparse.Namespace) -> None:
     if not results:
         print("No results collected.")
         return
     print("\n=== Benchmark Results ===")
     for result in results:
-        scenario = get_scenario(result.name)
-        print(f"\n[{result.name}] {scenario.description}")
+        print(f"\n[{result.name}] {get_scenario(result.name).description}")
         for key, value in result.metrics.items():
-            print(f"  {key}: {value:.6f}" if isinstance(value, float)
-                  else f"  {key}: {value}")
+            shown = f"{value:.6f}" if isinstance(value, float) else str(value)
+            print(f"  {key}: {shown}")
     if args.output:
         args.output.parent.mkdir(parents=True, exist_ok=True)
-        report = {
+        args.output.write_text(json.dumps({
             "timestamp": time.time(),
             "transport": "starway_amd-native (tcp + xgmi/hipipc)",
             "device": args.device or "cpu",
             "scenarios": [r.to_dict(include_samples=args.store_trace)
                           for r in results],
-        }
-        args.output.write_text(json.dumps(report, indent=2))
+        }, indent=2))
         print(f"\nJSON results written to {args.output}")
 
 
@@ -339,15 +327,10 @@ def main(argv: Sequence[str] | None = None) -> int:
     if args.role == "server":
         asyncio.run(run_server(args))
         return 0
-    if args.role == "client":
-        results = asyncio.run(run_client(args))
-        dump_results(results, args)
-        return 0
-    if args.role == "loopback":
-        results = asyncio.run(run_loopback(args))
-        dump_results(results, args)
-        return 0
-    raise ValueError(f"Unknown role {args.role}")
+    runner = run_client if args.role == "client" else run_loopback
+    results = asyncio.run(runner(args))
+    dump_results(results, args)
+    return 0
 
 
 if __name__ == "__main__":  # pragma: no cover
