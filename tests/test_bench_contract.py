@@ -49,3 +49,64 @@ def test_bench_two_rank_cpu_torchrun():
     assert d["config"]["endpoints"] == 2
     assert d["value"] > 0
     assert d["config"]["pingpong_64B_half_rtt_us"] > 0
+
+
+def test_bench_cli_two_process_roles(tmp_path):
+    """role=server + role=client in separate processes over localhost
+    (socket mode), one scenario, JSON report written by the client."""
+    port = "18441"
+    srv = subprocess.Popen(
+        [sys.executable, "-m", "starway_amd.bench", "--role", "server",
+         "--addr", "127.0.0.1", "--port", port],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        out_file = tmp_path / "r.json"
+        cli = subprocess.run(
+            [sys.executable, "-m", "starway_amd.bench", "--role", "client",
+             "--server-host", "127.0.0.1", "--port", port,
+             "--scenarios", "pingpong-flag", "--flag-iterations", "50",
+             "--flag-warmup", "5", "--output", str(out_file)],
+            capture_output=True, text=True, timeout=120, cwd=REPO)
+        assert cli.returncode == 0, cli.stdout[-800:] + cli.stderr[-400:]
+        report = json.loads(out_file.read_text())
+        assert report["scenarios"][0]["metrics"]["avg_rtt_us"] > 0
+        assert srv.wait(timeout=60) == 0
+    finally:
+        if srv.poll() is None:
+            srv.kill()
+            srv.wait()
+
+
+def test_bench_cli_worker_address_mode(tmp_path):
+    """Hex worker-address handshake through the CLI (reference C20
+    surface): server prints the blob, client connects with it."""
+    srv = subprocess.Popen(
+        [sys.executable, "-u", "-m", "starway_amd.bench", "--role",
+         "server", "--listen-mode", "worker"],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        import time as _t
+        blob_hex = None
+        deadline = _t.time() + 60
+        while _t.time() < deadline and blob_hex is None:
+            line = srv.stdout.readline()
+            if "worker address:" in line:
+                blob_hex = line.rsplit(":", 1)[1].strip()
+        assert blob_hex, "server never printed its worker address"
+        out_file = tmp_path / "r.json"
+        cli = subprocess.run(
+            [sys.executable, "-m", "starway_amd.bench", "--role", "client",
+             "--connect-mode", "worker", "--worker-address", blob_hex,
+             "--scenarios", "small-messages", "--small-iterations", "3",
+             "--small-warmup", "1", "--output", str(out_file)],
+            capture_output=True, text=True, timeout=120, cwd=REPO)
+        assert cli.returncode == 0, cli.stdout[-800:] + cli.stderr[-400:]
+        report = json.loads(out_file.read_text())
+        assert report["scenarios"][0]["metrics"]["messages_per_second"] > 0
+        assert srv.wait(timeout=60) == 0
+    finally:
+        if srv.poll() is None:
+            srv.kill()
+            srv.wait()
