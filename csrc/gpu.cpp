@@ -10,6 +10,7 @@
 #include "core.hpp"
 
 #include <dlfcn.h>
+#include <sys/stat.h>
 #include <hip/hip_runtime.h>
 
 #include <array>
@@ -192,12 +193,11 @@ bool calibrate(bool force, std::string* err) {
   g_calib.same_gbps = same;
   if (xgmi > 0) g_calib.xgmi_gbps = xgmi;
   std::string path = calib_path();
-  auto slash = path.rfind('/');
-  if (slash != std::string::npos) {
-    std::string dir = path.substr(0, slash);
-    std::string cmd = "mkdir -p '" + dir + "'";
-    int rc = system(cmd.c_str());
-    (void)rc;
+  // mkdir -p equivalent (no shell: the path comes from environment vars).
+  for (size_t pos = 1; (pos = path.find('/', pos)) != std::string::npos;
+       pos++) {
+    std::string dir = path.substr(0, pos);
+    if (!dir.empty()) mkdir(dir.c_str(), 0755);
   }
   if (FILE* f = fopen(path.c_str(), "w")) {
     fprintf(f, "same_gpu_gbps %.1f\n", g_calib.same_gbps);
