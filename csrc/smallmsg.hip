@@ -341,6 +341,35 @@ bool inbox_create(InboxInfo* out, int device, std::string* err) {
   out->slots = slots;
   out->slot_bytes = slot_bytes;
   memcpy(out->handle, &h, kIpcHandleBytes);
+  // Pre-warm this translation unit's code object (lazy hipModule load
+  // costs ~10-17 ms on first launch — pay it here at connection setup,
+  // not on the first message). Benign launches: seq 0 is the empty
+  // sentinel, sizes are 0, and the scratch flag cell is pool-local.
+  init_pools();
+  if (uint8_t* cell = g_flag_pool.get()) {
+    hipStream_t s = sm_stream(dev, SmStream::Push, 0);
+    PushArgs pa{};
+    pa.n = 1;
+    pa.d[0] = PushDesc{(const uint8_t*)base, (uint8_t*)base, 0, 0, 0};
+    hipLaunchKernelGGL(k_inbox_push, dim3(1), dim3(256), 0, s, pa);
+    UnpackArgs ua{};
+    ua.n = 1;
+    ua.results = (unsigned long long*)cell;
+    ua.spin_iters = 1;
+    ua.d[0] = UnpackDesc{(const uint8_t*)base, (uint8_t*)base, 0, 0};
+    hipLaunchKernelGGL(k_inbox_unpack, dim3(1), dim3(256), 0, s, ua);
+    ArmArgs aa{};
+    aa.slot = (const uint8_t*)base;
+    aa.expect_seq = ~0ull;  // never matches: expires after one spin
+    aa.dst = (uint8_t*)base;
+    aa.result = (unsigned long long*)cell;
+    aa.cancel = (unsigned int*)(cell + 8);
+    aa.spin_iters = 1;
+    hipLaunchKernelGGL(k_inbox_wait, dim3(1), dim3(256), 0, s, aa);
+    hipStreamSynchronize(s);
+    (void)hipGetLastError();
+    g_flag_pool.put(cell);
+  }
   return true;
 }
 
