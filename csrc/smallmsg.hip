@@ -19,7 +19,7 @@
 //     tag-checks the header, copies into the posted recv buffer and
 //     signals a pinned-host flag the progress thread polls — the receiver
 //     side then costs ~1 kernel-resident copy instead of launch+event
-//   * every spin is iteration-bounded (doorbell ~5 ms, unpack ~20 ms) and
+//   * every spin is iteration-bounded (order 1-6 ms of device time) and
 //     host-cancelable via a pinned cancel word, so no kernel can wedge the
 //     GPU (re-arming is the engine's job)
 //
@@ -487,7 +487,7 @@ void* inbox_unpack(const InboxInfo& mine, const UnpackMsg* msgs, int n,
     return nullptr;
   }
   static const unsigned spin =
-      (unsigned)env_u64("STARWAY_UNPACK_SPIN", 30000);  // ~20 ms bound
+      (unsigned)env_u64("STARWAY_UNPACK_SPIN", 30000);  // ~6 ms bound
 
   UnpackArgs args;
   args.n = n;
@@ -579,7 +579,7 @@ void* arm_recv(const InboxInfo& mine, uint64_t expect_seq, uint64_t tag,
   t->cancel = (unsigned int*)(cell + 8);
   t->device = mine.device;
   static const unsigned spin =
-      (unsigned)env_u64("STARWAY_ARM_SPIN", 8000);  // ~5 ms bound
+      (unsigned)env_u64("STARWAY_ARM_SPIN", 8000);  // ~1 ms bound
   ArmArgs a;
   a.slot = (const uint8_t*)(uintptr_t)mine.base +
            (expect_seq % mine.slots) * mine.slot_bytes;
