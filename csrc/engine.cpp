@@ -931,11 +931,16 @@ void Engine::start_send(Op* op, Connection* c) {
     // the moment `await asend` returns; the receiver pulls the snapshot.
     // On allocation failure fall back to zero-copy from the live buffer
     // (keepalive pins it; contents then must not change until flush).
-    if (op->capture.alloc(op->buf.size)) {
+    static const uint64_t kCaptureMax =
+        env_u64("STARWAY_CAPTURE_MAX", 1ull << 30);
+    if (op->buf.size <= kCaptureMax && op->capture.alloc(op->buf.size)) {
       memcpy(op->capture.data(), op->buf.ptr, op->buf.size);
       desc.addr = (uint64_t)(uintptr_t)op->capture.data();
       dead_objs_.push_back(std::move(op->keepalive));
     } else {
+      // Above the capture ceiling (or under memory pressure): the
+      // receiver pulls from the live buffer; the keepalive pins it and
+      // the caller must not modify it before flush (documented).
       desc.addr = (uint64_t)(uintptr_t)op->buf.ptr;
     }
     enqueue_frame(c, FT_RTS_CPU, op->tag, op->id, op->buf.size, &desc,
@@ -2731,6 +2736,16 @@ bool Engine::enqueue_eager(Connection* c, Op* op) {
       size_t off = 0;
       if (&back == &c->txq.front() && c->tx_front_written > back.head.size())
         off = c->tx_front_written - back.head.size();
+      // Capture ceiling (STARWAY_CAPTURE_MAX, default 1 GiB): above it a
+      // snapshot would double resident memory, so completion defers to
+      // write-out instead (still reuse-safe — the buffer is pinned and
+      // the caller's await resolves only once the bytes left).
+      static const uint64_t kCaptureMax =
+          env_u64("STARWAY_CAPTURE_MAX", 1ull << 30);
+      if (back.ext_len - off > kCaptureMax) {
+        back.owner = op;
+        return false;
+      }
       if (back.ext_own.alloc(back.ext_len)) {
         memcpy(back.ext_own.data() + off, back.ext + off, back.ext_len - off);
         back.ext = back.ext_own.data();
