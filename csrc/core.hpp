@@ -474,7 +474,7 @@ class Engine {
   void poll_gpu(bool& did_work);
   void enqueue_frame(Connection* c, FrameType t, uint64_t tag, uint64_t op_id,
                      uint64_t aux, const void* payload, size_t payload_len,
-                     bool priority);
+                     bool priority, uint16_t flags = 0);
   // Returns true when the payload is captured (written out or copied into
   // engine-owned memory) => the op may complete immediately. False => the
   // TxItem owns the op (deferred completion at write-out).
@@ -594,6 +594,7 @@ class Engine {
     std::vector<PendingUnpack> msgs;
     std::vector<bool> done;
     size_t remaining = 0;
+    bool probe = false;  // inbox bring-up probe: no recv op attached
   };
   std::vector<std::unique_ptr<UnpackBatch>> unpack_batches_;
   // Doorbell: at most one pre-armed wait kernel per engine.

@@ -1243,10 +1243,42 @@ void Engine::on_frame(Connection* c) {
       begin_eager(c);
       return;
     case FT_SMSG:
+      if (h.flags & 1) {
+        // Inbox bring-up probe: verify the seq word actually landed in
+        // our ring (one unpack of zero bytes), then acknowledge with a
+        // CREDIT so the sender activates the plane. Failure = plane
+        // stays off for this connection (fail-closed), loudly.
+        c->inbox_seen_seq = h.op_id;
+        gpu::UnpackMsg um{h.op_id, 0, nullptr};
+        std::string err;
+        void* t =
+            gpu::inbox_unpack(c->inbox_l, &um, 1, engine_lane_, &err);
+        if (t) {
+          auto batch = std::make_unique<UnpackBatch>();
+          batch->ticket = t;
+          batch->conn = c;
+          batch->msgs.push_back(
+              PendingUnpack{c, h.op_id, 0, 0, nullptr, nullptr});
+          batch->done.assign(1, false);
+          batch->remaining = 1;
+          batch->probe = true;
+          unpack_batches_.push_back(std::move(batch));
+        } else {
+          fprintf(stderr, "[starway] inbox probe unpack failed: %s\n",
+                  err.c_str());
+        }
+        return;
+      }
       handle_smsg(c, h.tag, h.aux, h.op_id);
       return;
     case FT_INBOX_CREDIT:
       if (h.aux > c->inbox_credit_base) c->inbox_credit_base = h.aux;
+      if (!c->inbox_r_active && c->inbox_r.slots &&
+          c->inbox_credit_base >= 1) {
+        // Bring-up probe acknowledged: the push plane is proven live.
+        c->inbox_r_active = true;
+        if (c->ep) c->ep->transports.emplace_back("xgmi", "inbox_push");
+      }
       return;
     case FT_HELLO:
     case FT_RTS:
@@ -1367,10 +1399,29 @@ void Engine::on_frame_payload(Connection* c) {
       memcpy(&c->inbox_r, c->rx_small.data(), sizeof(gpu::InboxInfo));
       if (c->inbox_r.slots && c->inbox_r.slot_bytes > gpu::kInboxHdrBytes &&
           gpu::available()) {
-        c->inbox_r_active = true;
-        c->inbox_next_seq = 1;
+        // FAIL-CLOSED bring-up: the ring activates only after a seq-1
+        // probe round-trips (push over xGMI -> peer unpack -> CREDIT).
+        // If the cross-device path misbehaves, the connection simply
+        // keeps using the proven RTS rendezvous plane.
+        c->inbox_r_active = false;
+        c->inbox_next_seq = 2;  // seq 1 is the probe
         c->inbox_credit_base = 0;
-        if (c->ep) c->ep->transports.emplace_back("xgmi", "inbox_push");
+        bool same_proc = memcmp(c->peer.uuid, process_uuid(), 16) == 0;
+        int run_dev = preferred_device_ >= 0 ? preferred_device_ : 0;
+        gpu::PushMsg probe{nullptr, 0, 1, 0};
+        std::string err;
+        void* ticket = gpu::inbox_push(c->inbox_r, same_proc, run_dev,
+                                       &probe, 1, engine_lane_, &err);
+        if (ticket) {
+          auto batch = std::make_unique<PushBatch>();
+          batch->ticket = ticket;
+          batch->conn = c;  // no ops: progress_pushes just frees it
+          push_batches_.push_back(std::move(batch));
+          enqueue_frame(c, FT_SMSG, 0, 1, 0, nullptr, 0, false,
+                        /*flags=*/1);
+        } else {
+          SW_DBG("inbox probe push failed: %s", err.c_str());
+        }
       }
       break;
     }
@@ -2011,6 +2062,19 @@ void Engine::progress_unpacks(bool& did_work) {
       b->done[k] = true;
       b->remaining--;
       PendingUnpack& m = b->msgs[k];
+      if (b->probe) {
+        if (r > 0 && m.conn && !m.conn->dead) {
+          m.conn->inbox_consumed = m.seq;
+          m.conn->inbox_credited = m.seq;
+          enqueue_frame(m.conn, FT_INBOX_CREDIT, 0, 0, m.seq, nullptr, 0,
+                        true);
+        } else if (r < 0) {
+          fprintf(stderr,
+                  "[starway] inbox bring-up probe never landed — push "
+                  "plane disabled for this connection (RTS fallback)\n");
+        }
+        continue;
+      }
       if (r < 0) {
         // The in-kernel wait expired before the payload landed. Under a
         // loaded launch queue the push kernel can start tens of ms late,
@@ -2658,13 +2722,15 @@ void Engine::complete_recv_from_unexpected(Op* op, UnexpectedMsg* um) {
 
 void Engine::enqueue_frame(Connection* c, FrameType t, uint64_t tag,
                            uint64_t op_id, uint64_t aux, const void* payload,
-                           size_t payload_len, bool priority) {
+                           size_t payload_len, bool priority,
+                           uint16_t flags) {
   if (c->dead) return;
   TxItem item;
   item.head.resize(sizeof(FrameHeader) + payload_len);
   FrameHeader h{};
   h.magic = kMagic;
   h.type = t;
+  h.flags = flags;
   h.tag = tag;
   h.size = payload_len;
   h.op_id = op_id;

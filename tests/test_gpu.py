@@ -670,6 +670,7 @@ def _inbox_child_client(port: int):
     async def inner():
         client = sw.Client()
         await client.aconnect("127.0.0.1", port)
+        await asyncio.sleep(0.3)  # let the ring's bring-up probe activate
         for i in range(16):
             src = torch.full((1024,), i, dtype=torch.uint8, device="cuda")
             torch.cuda.synchronize()
@@ -749,6 +750,7 @@ def _inbox_flush_close_sender(port: int):
     async def inner():
         client = sw.Client()
         await client.aconnect("127.0.0.1", port)
+        await asyncio.sleep(0.3)  # let the ring's bring-up probe activate
         for i in range(8):
             src = torch.full((512,), 40 + i, dtype=torch.uint8,
                              device="cuda")
