@@ -311,6 +311,7 @@ def main() -> int:
             # slower than lanes=1). The real 1-process-per-GPU topology
             # keeps the default 8 lanes.
             os.environ.setdefault("STARWAY_LANES", "1")
+            os.environ.setdefault("STARWAY_SM_LANES", "1")
 
     dist = None
     if world > 1:

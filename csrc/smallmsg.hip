@@ -232,9 +232,23 @@ enum class SmStream { Push, Unpack, Wait };
 // bound, and a second engine's doorbell queued behind it on the same
 // stream cannot start, serializing bidirectional latency paths into the
 // expiry cadence (measured: pingpong-flag RTT 116 us median, 5 ms tail).
+// STARWAY_SM_LANES caps the per-kind stream fan-out (power of two, max
+// 8). Shared-GPU emulation (many engines in few processes) sets 1 so the
+// stream count stays under the hardware-queue budget; the production
+// 1-process-per-GPU topology keeps distinct per-engine lanes.
+static int sm_lanes() {
+  static const int n = [] {
+    int v = (int)env_u64("STARWAY_SM_LANES", 8);
+    int p2 = 1;
+    while (p2 < v && p2 < 8) p2 <<= 1;
+    return p2;
+  }();
+  return n;
+}
+
 static hipStream_t sm_stream(int device, SmStream kind, int lane) {
   static std::map<std::tuple<int, int, int>, hipStream_t> streams;  // sm_mu
-  auto key = std::make_tuple(device, (int)kind, lane & 7);
+  auto key = std::make_tuple(device, (int)kind, lane & (sm_lanes() - 1));
   auto it = streams.find(key);
   if (it != streams.end()) return it->second;
   int prev;
