@@ -89,3 +89,63 @@ def test_matching_against_model(case):
             await server.aclose()
 
     asyncio.run(run())
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(scenario())
+def test_posted_recv_fifo_against_model(case):
+    """Dual of the test above: recvs are PRE-POSTED, messages arrive one
+    at a time. Model: an arriving message is claimed by the FIRST posted
+    recv (in post order) whose (tag, mask) accepts it; recvs that never
+    match fail with "cancel" at close."""
+    send_tags, recvs = case
+
+    async def run():
+        import socket
+
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+
+        server = Server()
+        client = Client()
+        server.listen(SERVER_ADDR, port)
+        await client.aconnect(SERVER_ADDR, port)
+        try:
+            bufs = [np.zeros(1, dtype=np.uint8) for _ in recvs]
+            futs = [server.arecv(bufs[i], tag, mask)
+                    for i, (tag, mask) in enumerate(recvs)]
+            await asyncio.sleep(0.05)  # recvs reach the posted table
+
+            # Model assignment: message j -> first free matching recv.
+            taken: dict[int, int] = {}  # recv idx -> message idx
+            for j, mtag in enumerate(send_tags):
+                for i, (tag, mask) in enumerate(recvs):
+                    if i in taken:
+                        continue
+                    if (mtag & mask) == (tag & mask):
+                        taken[i] = j
+                        break
+
+            for j, tag in enumerate(send_tags):
+                await client.asend(np.array([j], dtype=np.uint8), tag)
+            await client.aflush()
+
+            for i, fut in enumerate(futs):
+                if i in taken:
+                    got_tag, ln = await asyncio.wait_for(fut, 10)
+                    j = taken[i]
+                    assert got_tag == send_tags[j] and int(bufs[i][0]) == j, (
+                        f"recv {i} {recvs[i]} got ({got_tag}, {bufs[i][0]}), "
+                        f"model says msg {j} tag {send_tags[j]}")
+        finally:
+            await client.aclose()
+            await server.aclose()  # cancels the never-matching recvs
+            for i, fut in enumerate(futs):
+                if i not in taken:
+                    with pytest.raises(Exception, match="cancel"):
+                        await asyncio.wait_for(fut, 10)
+
+    asyncio.run(run())
