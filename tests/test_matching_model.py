@@ -149,3 +149,40 @@ def test_posted_recv_fifo_against_model(case):
                         await asyncio.wait_for(fut, 10)
 
     asyncio.run(run())
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(st.sampled_from([1, 100, 4095, 4097, 65536]),
+                min_size=1, max_size=10))
+def test_per_sender_order_across_eager_planes(sizes):
+    """Messages from one sender with one tag must deliver in send order to
+    a stream of wildcard recvs even when consecutive messages take
+    different eager paths (inline <= 4096 < zero-copy/captured)."""
+
+    async def run():
+        import socket
+
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+
+        server = Server()
+        client = Client()
+        server.listen(SERVER_ADDR, port)
+        await client.aconnect(SERVER_ADDR, port)
+        try:
+            for j, n in enumerate(sizes):
+                await client.asend(np.full(n, j % 251, dtype=np.uint8), 9)
+            await client.aflush()
+            for j, n in enumerate(sizes):
+                buf = np.zeros(max(sizes), dtype=np.uint8)
+                tag, ln = await asyncio.wait_for(server.arecv(buf, 0, 0), 10)
+                assert tag == 9 and ln == n, (j, n, ln)
+                assert (buf[:ln] == j % 251).all(), (j, n)
+        finally:
+            await client.aclose()
+            await server.aclose()
+
+    asyncio.run(run())
