@@ -235,3 +235,57 @@ async def test_close_cancels_window_deferred_sends(port):
     await client.asend(buf, 1)
     await client.aclose()
     await server.aclose()
+
+
+def _window_doomed_sender(port, q):
+    os.environ["STARWAY_SEND_WINDOW"] = "1M"
+
+    async def inner():
+        client = Client()
+        await client.aconnect(ADDR, port)
+        # One admitted CMA rendezvous + two window-deferred behind it.
+        # Deferred sends only complete once admitted, so do NOT await
+        # them individually — their futures must FAIL at peer death.
+        send_futs = [client.asend(_pattern(2 << 20, 30 + i), 20 + i)
+                     for i in range(3)]
+        flush_fut = client.aflush()
+        q.put("sent")
+        try:
+            await asyncio.wait_for(flush_fut, timeout=20)
+            q.put("flush-completed")
+        except Exception as exc:
+            q.put(f"flush-failed:{exc}")
+        outcomes = await asyncio.gather(*send_futs, return_exceptions=True)
+        failed = sum(isinstance(o, Exception) for o in outcomes)
+        q.put(f"send-failures:{failed}")
+        await client.aclose()
+        q.put("closed")
+
+    asyncio.run(inner())
+
+
+async def test_deferred_sends_fail_when_peer_dies(port):
+    """Peer death with window-DEFERRED rendezvous sends queued: the flush
+    covering them must fail (not hang) and close must still succeed."""
+    server = Server()
+    server.listen(ADDR, port)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_window_doomed_sender, args=(port, q))
+    p.start()
+    try:
+        assert q.get(timeout=30) == "sent"
+        await asyncio.sleep(0.2)
+        await server.aclose()  # peer disappears mid-rendezvous
+        verdict = q.get(timeout=30)
+        assert verdict.startswith("flush-failed"), verdict
+        assert "reset" in verdict or "closed" in verdict or "cancel" in verdict
+        failures = q.get(timeout=30)
+        assert failures == "send-failures:2", failures  # the deferred pair
+        assert q.get(timeout=30) == "closed"
+    finally:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.kill()
+            p.join()
+        p.close()
