@@ -502,6 +502,10 @@ async def test_inbox_small_roundtrip(nbytes):
                             device="cuda")
         dst = torch.zeros_like(src)
         torch.cuda.synchronize()
+        # Cold start: the ring's bring-up probe includes the first kernel
+        # module load (~15 ms); settle so the stats assertion below sees
+        # the activated plane even when this test runs first.
+        await asyncio.sleep(0.3)
         fut = server.arecv(dst, 0, 0)
         await asyncio.sleep(0.01)
         await client.asend(src, 5)
@@ -537,6 +541,7 @@ async def test_inbox_many_concurrent():
         dsts = [torch.zeros(1024, dtype=torch.uint8, device="cuda")
                 for _ in range(n)]
         torch.cuda.synchronize()
+        await asyncio.sleep(0.3)  # cold-start probe settle (stats assert)
         recvs = [server.arecv(dsts[i], 0, 0) for i in range(n)]
         await asyncio.sleep(0.01)
         await asyncio.gather(*(client.asend(srcs[i], 100 + i)
@@ -620,6 +625,7 @@ def _doorbell_child(q):
 async def _doorbell_body():
     async with loopback() as (server, client):
         ep = next(iter(server.list_clients()))
+        await asyncio.sleep(0.3)  # cold-start probe settle (stats assert)
         ping = torch.full((64,), 7, dtype=torch.uint8, device="cuda")
         pong = torch.full((64,), 9, dtype=torch.uint8, device="cuda")
         rx_s = torch.zeros(64, dtype=torch.uint8, device="cuda")
